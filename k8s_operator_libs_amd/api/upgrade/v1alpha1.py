@@ -1,0 +1,122 @@
+"""Driver-upgrade policy API types (group ``upgrade.amd.com``, v1alpha1).
+
+Capability parity with the reference's
+``api/upgrade/v1alpha1/upgrade_spec.go:27-110``: the four spec fragments a
+consuming operator embeds in its own CRD to configure the rolling
+driver-upgrade state machine.  Field names, defaults and validation semantics
+match the reference's kubebuilder markers:
+
+- ``DriverUpgradePolicySpec.auto_upgrade`` default False (spec.go:32)
+- ``max_parallel_upgrades`` default 1, minimum 0, 0 = unlimited (spec.go:33-38)
+- ``max_unavailable`` int-or-percent, default "25%" (spec.go:39-45)
+- ``WaitForCompletionSpec`` pod_selector/timeout_seconds (spec.go:52-64)
+- ``PodDeletionSpec`` force/timeout 300 s/delete_emptydir_data (spec.go:67-83)
+- ``DrainSpec`` enable/force/pod_selector/timeout 300 s/delete_emptydir_data
+  (spec.go:86-110)
+
+Pydantic replaces Go's kubebuilder validation + generated deepcopy: models are
+validated on construction, ``model_copy(deep=True)`` is the DeepCopy
+equivalent, and the JSON aliases are the exact wire names used inside a CRD
+(``autoUpgrade``, ``maxParallelUpgrades``, ...).
+"""
+
+from __future__ import annotations
+
+import math
+import re
+from typing import Optional, Union
+
+from pydantic import BaseModel, ConfigDict, Field, field_validator
+
+_PERCENT_RE = re.compile(r"^(\d+)%$")
+
+
+class IntOrString:
+    """Kubernetes ``intstr.IntOrString`` semantics for int-or-percent fields."""
+
+    @staticmethod
+    def scaled_value(value: Union[int, str, None], total: int, round_up: bool) -> int:
+        """Mirror of apimachinery ``intstr.GetScaledValueFromIntOrPercent``.
+
+        An int (or int-like string) is returned as-is; ``"25%"`` scales against
+        ``total`` and rounds up or down per ``round_up``.  The in-place mode
+        rounds **up** when computing maxUnavailable (upgrade_inplace.go:54-60).
+        """
+        if value is None:
+            return 0
+        if isinstance(value, int):
+            return value
+        value = value.strip()
+        m = _PERCENT_RE.match(value)
+        if m:
+            pct = int(m.group(1))
+            exact = pct * total / 100.0
+            return math.ceil(exact) if round_up else math.floor(exact)
+        return int(value)
+
+    @staticmethod
+    def validate(value: Union[int, str, None]) -> Union[int, str, None]:
+        if value is None or isinstance(value, int):
+            return value
+        if _PERCENT_RE.match(value.strip()):
+            return value.strip()
+        return int(value)  # raises ValueError on junk
+
+
+class _SpecBase(BaseModel):
+    model_config = ConfigDict(populate_by_name=True, extra="forbid")
+
+    def deep_copy(self):
+        """DeepCopy equivalent of zz_generated.deepcopy.go."""
+        return self.model_copy(deep=True)
+
+
+class WaitForCompletionSpec(_SpecBase):
+    """Wait for selected workload pods to complete before deletion
+    (upgrade_spec.go:52-64)."""
+
+    # Label selector (string form, e.g. "app=training-job") of pods to wait on.
+    pod_selector: str = Field(default="", alias="podSelector")
+    # 0 means wait forever.
+    timeout_seconds: int = Field(default=0, ge=0, alias="timeoutSecond")
+
+
+class PodDeletionSpec(_SpecBase):
+    """Controlled deletion of selected pods before driver restart
+    (upgrade_spec.go:67-83)."""
+
+    force: bool = Field(default=False)
+    timeout_seconds: int = Field(default=300, ge=0, alias="timeoutSecond")
+    delete_emptydir_data: bool = Field(default=False, alias="deleteEmptyDir")
+
+
+class DrainSpec(_SpecBase):
+    """Full node drain configuration (upgrade_spec.go:86-110)."""
+
+    enable: bool = Field(default=False)
+    force: bool = Field(default=False)
+    pod_selector: str = Field(default="", alias="podSelector")
+    timeout_seconds: int = Field(default=300, ge=0, alias="timeoutSeconds")
+    delete_emptydir_data: bool = Field(default=False, alias="deleteEmptyDir")
+
+
+class DriverUpgradePolicySpec(_SpecBase):
+    """Top-level driver upgrade policy (upgrade_spec.go:27-50)."""
+
+    auto_upgrade: bool = Field(default=False, alias="autoUpgrade")
+    # 0 = no limit on concurrently upgrading nodes.
+    max_parallel_upgrades: int = Field(default=1, ge=0, alias="maxParallelUpgrades")
+    # int or percent string; percent is relative to total managed nodes.
+    max_unavailable: Optional[Union[int, str]] = Field(
+        default="25%", alias="maxUnavailable"
+    )
+    wait_for_completion: Optional[WaitForCompletionSpec] = Field(
+        default=None, alias="waitForCompletion"
+    )
+    pod_deletion: Optional[PodDeletionSpec] = Field(default=None, alias="podDeletion")
+    drain_spec: Optional[DrainSpec] = Field(default=None, alias="drainSpec")
+
+    @field_validator("max_unavailable")
+    @classmethod
+    def _validate_max_unavailable(cls, v):
+        return IntOrString.validate(v)

@@ -1,0 +1,39 @@
+"""Typed Kubernetes API errors (apimachinery ``k8s.io/apimachinery/pkg/api/errors`` analogue)."""
+
+from __future__ import annotations
+
+
+class ApiError(Exception):
+    """Base class; carries an HTTP-style status code."""
+
+    code = 500
+
+    def __init__(self, message: str = "") -> None:
+        super().__init__(message or self.__class__.__name__)
+        self.message = message
+
+
+class NotFoundError(ApiError):
+    code = 404
+
+
+class AlreadyExistsError(ApiError):
+    code = 409
+
+
+class ConflictError(ApiError):
+    """Optimistic-concurrency failure (stale resourceVersion)."""
+
+    code = 409
+
+
+class BadRequestError(ApiError):
+    code = 400
+
+
+def is_not_found(exc: BaseException) -> bool:
+    return isinstance(exc, NotFoundError)
+
+
+def is_conflict(exc: BaseException) -> bool:
+    return isinstance(exc, ConflictError)

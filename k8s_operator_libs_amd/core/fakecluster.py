@@ -1,0 +1,345 @@
+"""In-memory kube-apiserver: the envtest equivalent.
+
+The reference tests against envtest — a real kube-apiserver + etcd with no
+kubelet/scheduler/controllers.  This module provides the same
+substrate natively: typed CRUD with resourceVersions and optimistic
+concurrency, label/field-selector LISTs, JSON merge patches (with
+optimistic-lock support), finalizer-aware deletion, pod eviction, and watch
+streams.  Nothing "runs" pods: like envtest, tests and benchmarks force pod /
+DaemonSet statuses directly.
+
+It is also the storage engine of :mod:`k8s_operator_libs_amd.core.apiserver`,
+which serves it over HTTP so the REST client can be tested wire-level.
+"""
+
+from __future__ import annotations
+
+import itertools
+import queue
+import threading
+import time
+import uuid
+from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+
+from . import meta
+from .errors import (
+    AlreadyExistsError,
+    BadRequestError,
+    ConflictError,
+    NotFoundError,
+)
+from .meta import FieldSelector, K8sObject, LabelSelector
+
+# (apiVersion, kind) -> (plural, namespaced)
+_BUILTIN_KINDS: Dict[Tuple[str, str], Tuple[str, bool]] = {
+    ("v1", "Node"): ("nodes", False),
+    ("v1", "Pod"): ("pods", True),
+    ("v1", "Event"): ("events", True),
+    ("v1", "Namespace"): ("namespaces", False),
+    ("apps/v1", "DaemonSet"): ("daemonsets", True),
+    ("apps/v1", "ControllerRevision"): ("controllerrevisions", True),
+    ("apiextensions.k8s.io/v1", "CustomResourceDefinition"): (
+        "customresourcedefinitions",
+        False,
+    ),
+    # AMD maintenance-operator API (requestor mode; the reference's analogue
+    # is maintenance.nvidia.com/v1alpha1 NodeMaintenance).
+    ("maintenance.amd.com/v1alpha1", "NodeMaintenance"): ("nodemaintenances", True),
+}
+
+
+class Watch:
+    """A watch stream: a queue of ``(event_type, object)`` tuples where
+    event_type is ``ADDED`` / ``MODIFIED`` / ``DELETED``."""
+
+    def __init__(self, cluster: "FakeCluster", key: Tuple[str, str]) -> None:
+        self._cluster = cluster
+        self._key = key
+        self.events: "queue.Queue[Tuple[str, K8sObject]]" = queue.Queue()
+        self._stopped = False
+
+    def next(self, timeout: Optional[float] = None) -> Optional[Tuple[str, K8sObject]]:
+        try:
+            return self.events.get(timeout=timeout)
+        except queue.Empty:
+            return None
+
+    def stop(self) -> None:
+        self._stopped = True
+        self._cluster._remove_watch(self)
+
+
+class FakeCluster:
+    """Thread-safe in-memory Kubernetes object store."""
+
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        # (apiVersion, kind) -> {(namespace, name): object}
+        self._store: Dict[Tuple[str, str], Dict[Tuple[str, str], K8sObject]] = {}
+        self._rv = itertools.count(1)
+        self._kinds = dict(_BUILTIN_KINDS)
+        self._watches: Dict[Tuple[str, str], List[Watch]] = {}
+        self._change_hooks: List[Callable[[str, K8sObject], None]] = []
+
+    # -- kind registry -------------------------------------------------------
+
+    def register_kind(self, api_version: str, kind: str, plural: str, namespaced: bool) -> None:
+        with self._lock:
+            self._kinds[(api_version, kind)] = (plural, namespaced)
+
+    def register_crd(self, crd: K8sObject) -> None:
+        """Make a created CustomResourceDefinition's kind servable."""
+        spec = crd.get("spec", {})
+        group = spec.get("group", "")
+        names = spec.get("names", {})
+        kind = names.get("kind", "")
+        plural = names.get("plural", "")
+        namespaced = spec.get("scope", "Namespaced") == "Namespaced"
+        for ver in spec.get("versions", []):
+            if ver.get("served", True):
+                self.register_kind(f"{group}/{ver['name']}", kind, plural, namespaced)
+
+    def lookup_kind(self, api_version: str, kind: str) -> Tuple[str, bool]:
+        try:
+            return self._kinds[(api_version, kind)]
+        except KeyError:
+            raise BadRequestError(f"unknown kind {api_version}/{kind}") from None
+
+    def lookup_by_plural(self, api_version: str, plural: str) -> Optional[str]:
+        """Reverse lookup used by the HTTP server and CRD-establish polling."""
+        with self._lock:
+            for (av, kind), (pl, _ns) in self._kinds.items():
+                if av == api_version and pl == plural:
+                    return kind
+        return None
+
+    # -- internal helpers ----------------------------------------------------
+
+    def _next_rv(self) -> str:
+        return str(next(self._rv))
+
+    @staticmethod
+    def _obj_key(obj: K8sObject) -> Tuple[str, str]:
+        return (meta.api_version(obj), meta.kind(obj))
+
+    def _bucket(self, api_version: str, kind: str) -> Dict[Tuple[str, str], K8sObject]:
+        self.lookup_kind(api_version, kind)  # validate
+        return self._store.setdefault((api_version, kind), {})
+
+    def _notify(self, event_type: str, obj: K8sObject) -> None:
+        key = self._obj_key(obj)
+        snapshot = meta.deep_copy(obj)
+        for w in self._watches.get(key, []):
+            w.events.put((event_type, snapshot))
+        for hook in self._change_hooks:
+            hook(event_type, snapshot)
+
+    def add_change_hook(self, hook: Callable[[str, K8sObject], None]) -> None:
+        """Register a callback fired on every mutation (used by simulated
+        kubelets / maintenance operators in tests and benchmarks)."""
+        with self._lock:
+            self._change_hooks.append(hook)
+
+    def _remove_watch(self, watch: Watch) -> None:
+        with self._lock:
+            lst = self._watches.get(watch._key, [])
+            if watch in lst:
+                lst.remove(watch)
+
+    # -- CRUD ----------------------------------------------------------------
+
+    def create(self, obj: K8sObject) -> K8sObject:
+        obj = meta.deep_copy(obj)
+        api_version, kind = self._obj_key(obj)
+        with self._lock:
+            _, namespaced = self.lookup_kind(api_version, kind)
+            md = obj.setdefault("metadata", {})
+            ns = md.get("namespace", "")
+            if namespaced and not ns:
+                md["namespace"] = ns = "default"
+            if not namespaced:
+                md.pop("namespace", None)
+                ns = ""
+            name_ = md.get("name", "")
+            if not name_:
+                gen = md.get("generateName")
+                if not gen:
+                    raise BadRequestError("metadata.name is required")
+                name_ = f"{gen}{uuid.uuid4().hex[:6]}"
+                md["name"] = name_
+            bucket = self._bucket(api_version, kind)
+            if (ns, name_) in bucket:
+                raise AlreadyExistsError(f"{kind} {ns}/{name_} already exists")
+            md["uid"] = str(uuid.uuid4())
+            md["resourceVersion"] = self._next_rv()
+            md.setdefault("creationTimestamp", _now_iso())
+            md.setdefault("generation", 1)
+            bucket[(ns, name_)] = obj
+            if kind == "CustomResourceDefinition":
+                self.register_crd(obj)
+                self._establish_crd(obj)
+            self._notify("ADDED", obj)
+            return meta.deep_copy(obj)
+
+    def get(self, api_version: str, kind: str, name: str, namespace: str = "") -> K8sObject:
+        with self._lock:
+            _, namespaced = self.lookup_kind(api_version, kind)
+            ns = namespace if namespaced else ""
+            bucket = self._bucket(api_version, kind)
+            obj = bucket.get((ns, name))
+            if obj is None:
+                raise NotFoundError(f"{kind} {ns}/{name} not found")
+            return meta.deep_copy(obj)
+
+    def list(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: str = "",
+        field_selector: str = "",
+    ) -> List[K8sObject]:
+        lsel = LabelSelector(label_selector)
+        fsel = FieldSelector(field_selector)
+        with self._lock:
+            bucket = self._bucket(api_version, kind)
+            out = []
+            for (ns, _name), obj in bucket.items():
+                if namespace is not None and namespace != "" and ns != namespace:
+                    continue
+                if label_selector and not lsel.matches_object(obj):
+                    continue
+                if field_selector and not fsel.matches_object(obj):
+                    continue
+                out.append(meta.deep_copy(obj))
+            out.sort(key=lambda o: (meta.namespace(o), meta.name(o)))
+            return out
+
+    def update(self, obj: K8sObject) -> K8sObject:
+        obj = meta.deep_copy(obj)
+        api_version, kind = self._obj_key(obj)
+        with self._lock:
+            _, namespaced = self.lookup_kind(api_version, kind)
+            ns = meta.namespace(obj) if namespaced else ""
+            name_ = meta.name(obj)
+            bucket = self._bucket(api_version, kind)
+            stored = bucket.get((ns, name_))
+            if stored is None:
+                raise NotFoundError(f"{kind} {ns}/{name_} not found")
+            rv = meta.resource_version(obj)
+            if rv and rv != meta.resource_version(stored):
+                raise ConflictError(
+                    f"{kind} {ns}/{name_}: resourceVersion {rv} is stale"
+                )
+            # Immutable server-side fields carry over.
+            obj["metadata"]["uid"] = stored["metadata"]["uid"]
+            obj["metadata"]["creationTimestamp"] = stored["metadata"]["creationTimestamp"]
+            if "deletionTimestamp" in stored["metadata"]:
+                obj["metadata"]["deletionTimestamp"] = stored["metadata"]["deletionTimestamp"]
+            obj["metadata"]["resourceVersion"] = self._next_rv()
+            obj["metadata"]["generation"] = stored["metadata"].get("generation", 1) + 1
+            bucket[(ns, name_)] = obj
+            self._notify("MODIFIED", obj)
+            self._finalize_if_ready(api_version, kind, ns, name_)
+            return meta.deep_copy(bucket.get((ns, name_), obj))
+
+    def patch(
+        self,
+        api_version: str,
+        kind: str,
+        name: str,
+        patch: K8sObject,
+        namespace: str = "",
+    ) -> K8sObject:
+        """Apply an RFC 7386 JSON merge patch.  If the patch carries
+        ``metadata.resourceVersion`` it acts as an optimistic lock (the
+        shared-requestor protocol relies on this — reference
+        upgrade_requestor.go:320-368)."""
+        with self._lock:
+            _, namespaced = self.lookup_kind(api_version, kind)
+            ns = namespace if namespaced else ""
+            bucket = self._bucket(api_version, kind)
+            stored = bucket.get((ns, name))
+            if stored is None:
+                raise NotFoundError(f"{kind} {ns}/{name} not found")
+            patch_rv = (
+                patch.get("metadata", {}).get("resourceVersion")
+                if isinstance(patch.get("metadata"), dict)
+                else None
+            )
+            if patch_rv and patch_rv != meta.resource_version(stored):
+                raise ConflictError(
+                    f"{kind} {ns}/{name}: resourceVersion {patch_rv} is stale"
+                )
+            meta.json_merge_patch(stored, patch)
+            stored["metadata"]["name"] = name  # patches cannot rename
+            stored["metadata"]["resourceVersion"] = self._next_rv()
+            self._notify("MODIFIED", stored)
+            self._finalize_if_ready(api_version, kind, ns, name)
+            out = bucket.get((ns, name))
+            return meta.deep_copy(out) if out is not None else {"deleted": True}
+
+    def delete(self, api_version: str, kind: str, name: str, namespace: str = "") -> None:
+        with self._lock:
+            _, namespaced = self.lookup_kind(api_version, kind)
+            ns = namespace if namespaced else ""
+            bucket = self._bucket(api_version, kind)
+            stored = bucket.get((ns, name))
+            if stored is None:
+                raise NotFoundError(f"{kind} {ns}/{name} not found")
+            finalizers = stored.get("metadata", {}).get("finalizers") or []
+            if finalizers:
+                if "deletionTimestamp" not in stored["metadata"]:
+                    stored["metadata"]["deletionTimestamp"] = _now_iso()
+                    stored["metadata"]["resourceVersion"] = self._next_rv()
+                    self._notify("MODIFIED", stored)
+                return
+            del bucket[(ns, name)]
+            self._notify("DELETED", stored)
+
+    def _finalize_if_ready(self, api_version: str, kind: str, ns: str, name: str) -> None:
+        """Remove an object whose deletion was pending once finalizers empty."""
+        bucket = self._bucket(api_version, kind)
+        stored = bucket.get((ns, name))
+        if stored is None:
+            return
+        md = stored.get("metadata", {})
+        if "deletionTimestamp" in md and not (md.get("finalizers") or []):
+            del bucket[(ns, name)]
+            self._notify("DELETED", stored)
+
+    # -- pods ----------------------------------------------------------------
+
+    def evict_pod(self, name: str, namespace: str) -> None:
+        """Eviction API: like envtest there is no kubelet, so eviction is an
+        immediate graceful delete."""
+        self.delete("v1", "Pod", name, namespace)
+
+    # -- watches -------------------------------------------------------------
+
+    def watch(self, api_version: str, kind: str) -> Watch:
+        with self._lock:
+            self.lookup_kind(api_version, kind)
+            w = Watch(self, (api_version, kind))
+            self._watches.setdefault((api_version, kind), []).append(w)
+            return w
+
+    # -- CRD establishment ---------------------------------------------------
+
+    def _establish_crd(self, crd: K8sObject) -> None:
+        """A real apiserver establishes CRDs asynchronously; we mark the
+        Established condition immediately (discovery still exercises the
+        polling path through :meth:`lookup_by_plural`)."""
+        crd.setdefault("status", {})["conditions"] = [
+            {"type": "Established", "status": "True", "reason": "InitialNamesAccepted"}
+        ]
+
+    # -- convenience for tests/benchmarks ------------------------------------
+
+    def object_count(self) -> int:
+        with self._lock:
+            return sum(len(b) for b in self._store.values())
+
+
+def _now_iso() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())

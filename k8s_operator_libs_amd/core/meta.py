@@ -1,0 +1,230 @@
+"""Helpers over JSON-shaped (dict) Kubernetes objects.
+
+Objects travel through this library exactly as they travel on the wire: plain
+``dict`` trees (the controller-runtime "unstructured" style).  This module
+provides the metadata accessors, label/field-selector matching and
+JSON-merge-patch used by the client substrate and the upgrade managers.
+"""
+
+from __future__ import annotations
+
+import copy
+import re
+from typing import Any, Dict, Iterable, List, Optional
+
+K8sObject = Dict[str, Any]
+
+
+# -- metadata accessors ------------------------------------------------------
+
+def name(obj: K8sObject) -> str:
+    return obj.get("metadata", {}).get("name", "")
+
+
+def namespace(obj: K8sObject) -> str:
+    return obj.get("metadata", {}).get("namespace", "")
+
+
+def uid(obj: K8sObject) -> str:
+    return obj.get("metadata", {}).get("uid", "")
+
+
+def resource_version(obj: K8sObject) -> str:
+    return obj.get("metadata", {}).get("resourceVersion", "")
+
+
+def kind(obj: K8sObject) -> str:
+    return obj.get("kind", "")
+
+
+def api_version(obj: K8sObject) -> str:
+    return obj.get("apiVersion", "")
+
+
+def labels(obj: K8sObject) -> Dict[str, str]:
+    return obj.setdefault("metadata", {}).setdefault("labels", {})
+
+
+def annotations(obj: K8sObject) -> Dict[str, str]:
+    return obj.setdefault("metadata", {}).setdefault("annotations", {})
+
+
+def get_label(obj: K8sObject, key: str, default: str = "") -> str:
+    return obj.get("metadata", {}).get("labels", {}).get(key, default)
+
+
+def get_annotation(obj: K8sObject, key: str, default: str = "") -> str:
+    return obj.get("metadata", {}).get("annotations", {}).get(key, default)
+
+
+def owner_references(obj: K8sObject) -> List[Dict[str, Any]]:
+    return obj.get("metadata", {}).get("ownerReferences", []) or []
+
+
+def controller_owner(obj: K8sObject) -> Optional[Dict[str, Any]]:
+    """The ownerReference flagged controller=true, if any."""
+    for ref in owner_references(obj):
+        if ref.get("controller"):
+            return ref
+    return None
+
+
+def deep_copy(obj: K8sObject) -> K8sObject:
+    return copy.deepcopy(obj)
+
+
+def dotted_get(obj: K8sObject, path: str, default: Any = None) -> Any:
+    """Fetch ``spec.nodeName``-style dotted paths."""
+    cur: Any = obj
+    for part in path.split("."):
+        if not isinstance(cur, dict) or part not in cur:
+            return default
+        cur = cur[part]
+    return cur
+
+
+# -- label selectors ---------------------------------------------------------
+
+_IN_RE = re.compile(r"^\s*([A-Za-z0-9._/-]+)\s+(in|notin)\s+\(([^)]*)\)\s*$")
+
+
+class LabelSelector:
+    """String label selector with the semantics of apimachinery
+    ``labels.Parse``: comma-joined requirements of the forms ``k=v``,
+    ``k==v``, ``k!=v``, ``k in (a,b)``, ``k notin (a,b)``, ``k`` (exists) and
+    ``!k`` (not exists)."""
+
+    def __init__(self, selector: str = "") -> None:
+        self.raw = selector or ""
+        self._reqs = self._parse(self.raw)
+
+    @staticmethod
+    def _split_requirements(s: str) -> Iterable[str]:
+        # Commas inside "in (a,b)" parens are not separators.
+        depth = 0
+        cur = []
+        for ch in s:
+            if ch == "(":
+                depth += 1
+            elif ch == ")":
+                depth -= 1
+            if ch == "," and depth == 0:
+                yield "".join(cur)
+                cur = []
+            else:
+                cur.append(ch)
+        if cur:
+            yield "".join(cur)
+
+    @classmethod
+    def _parse(cls, s: str):
+        reqs = []
+        if not s.strip():
+            return reqs
+        for part in cls._split_requirements(s):
+            part = part.strip()
+            if not part:
+                continue
+            m = _IN_RE.match(part)
+            if m:
+                key, op, vals = m.groups()
+                values = {v.strip() for v in vals.split(",") if v.strip()}
+                reqs.append((key, op, values))
+            elif "!=" in part:
+                key, _, val = part.partition("!=")
+                reqs.append((key.strip(), "!=", val.strip()))
+            elif "==" in part:
+                key, _, val = part.partition("==")
+                reqs.append((key.strip(), "=", val.strip()))
+            elif "=" in part:
+                key, _, val = part.partition("=")
+                reqs.append((key.strip(), "=", val.strip()))
+            elif part.startswith("!"):
+                reqs.append((part[1:].strip(), "!exists", None))
+            else:
+                reqs.append((part, "exists", None))
+        return reqs
+
+    def matches(self, lbls: Dict[str, str]) -> bool:
+        lbls = lbls or {}
+        for key, op, val in self._reqs:
+            if op == "=":
+                if lbls.get(key) != val:
+                    return False
+            elif op == "!=":
+                # apimachinery: != also matches objects lacking the key
+                if key in lbls and lbls[key] == val:
+                    return False
+            elif op == "in":
+                if lbls.get(key) not in val:
+                    return False
+            elif op == "notin":
+                if key in lbls and lbls[key] in val:
+                    return False
+            elif op == "exists":
+                if key not in lbls:
+                    return False
+            elif op == "!exists":
+                if key in lbls:
+                    return False
+        return True
+
+    def matches_object(self, obj: K8sObject) -> bool:
+        return self.matches(obj.get("metadata", {}).get("labels", {}) or {})
+
+
+def match_labels_selector(match: Dict[str, str]) -> LabelSelector:
+    """Selector from a ``matchLabels`` map (DaemonSet spec.selector)."""
+    sel = LabelSelector("")
+    sel._reqs = [(k, "=", v) for k, v in (match or {}).items()]
+    sel.raw = ",".join(f"{k}={v}" for k, v in (match or {}).items())
+    return sel
+
+
+class FieldSelector:
+    """Equality-only field selector (``spec.nodeName=foo,status.phase=Running``)."""
+
+    def __init__(self, selector: str = "") -> None:
+        self.raw = selector or ""
+        self._reqs = []
+        for part in self.raw.split(","):
+            part = part.strip()
+            if not part:
+                continue
+            if "!=" in part:
+                key, _, val = part.partition("!=")
+                self._reqs.append((key.strip(), "!=", val.strip()))
+            else:
+                key, _, val = part.partition("=")
+                self._reqs.append((key.strip(), "=", val.strip()))
+
+    def matches_object(self, obj: K8sObject) -> bool:
+        for key, op, val in self._reqs:
+            # metadata.name / metadata.namespace are the common specials
+            actual = dotted_get(obj, key, "")
+            actual = "" if actual is None else str(actual)
+            if op == "=" and actual != val:
+                return False
+            if op == "!=" and actual == val:
+                return False
+        return True
+
+
+# -- JSON merge patch (RFC 7386) --------------------------------------------
+
+def json_merge_patch(target: Any, patch: Any) -> Any:
+    """Apply an RFC 7386 merge patch: dicts merge recursively, ``None``
+    deletes a key, everything else replaces.  Returns the patched value
+    (mutates dict targets in place)."""
+    if not isinstance(patch, dict):
+        return copy.deepcopy(patch)
+    if not isinstance(target, dict):
+        target = {}
+    for key, value in patch.items():
+        if value is None:
+            target.pop(key, None)
+        elif isinstance(value, dict):
+            target[key] = json_merge_patch(target.get(key), value)
+        else:
+            target[key] = copy.deepcopy(value)
+    return target

@@ -1,0 +1,87 @@
+"""Policy API type tests (reference api/upgrade/v1alpha1/upgrade_spec.go)."""
+
+import pytest
+from pydantic import ValidationError
+
+from k8s_operator_libs_amd.api.upgrade.v1alpha1 import (
+    DrainSpec,
+    DriverUpgradePolicySpec,
+    IntOrString,
+    PodDeletionSpec,
+    WaitForCompletionSpec,
+)
+
+
+def test_policy_defaults():
+    p = DriverUpgradePolicySpec()
+    assert p.auto_upgrade is False
+    assert p.max_parallel_upgrades == 1
+    assert p.max_unavailable == "25%"
+    assert p.wait_for_completion is None
+    assert p.pod_deletion is None
+    assert p.drain_spec is None
+
+
+def test_policy_wire_aliases():
+    p = DriverUpgradePolicySpec.model_validate(
+        {
+            "autoUpgrade": True,
+            "maxParallelUpgrades": 4,
+            "maxUnavailable": 3,
+            "drainSpec": {"enable": True, "timeoutSeconds": 60},
+            "podDeletion": {"force": True, "deleteEmptyDir": True},
+            "waitForCompletion": {"podSelector": "app=job", "timeoutSecond": 30},
+        }
+    )
+    assert p.max_parallel_upgrades == 4
+    assert p.max_unavailable == 3
+    assert p.drain_spec.enable and p.drain_spec.timeout_seconds == 60
+    assert p.pod_deletion.force and p.pod_deletion.delete_emptydir_data
+    assert p.wait_for_completion.pod_selector == "app=job"
+    # round-trip through the wire shape
+    wire = p.model_dump(by_alias=True)
+    assert wire["maxParallelUpgrades"] == 4
+    assert DriverUpgradePolicySpec.model_validate(wire) == p
+
+
+def test_nested_defaults_match_reference():
+    assert PodDeletionSpec().timeout_seconds == 300  # spec.go:72-77
+    assert DrainSpec().timeout_seconds == 300  # spec.go:100-104
+    assert DrainSpec().enable is False
+    assert WaitForCompletionSpec().timeout_seconds == 0
+
+
+def test_validation_rejects_negatives():
+    with pytest.raises(ValidationError):
+        DriverUpgradePolicySpec(maxParallelUpgrades=-1)
+    with pytest.raises(ValidationError):
+        DrainSpec(timeoutSeconds=-5)
+
+
+def test_max_unavailable_percent_validation():
+    assert DriverUpgradePolicySpec(maxUnavailable="50%").max_unavailable == "50%"
+    with pytest.raises(ValidationError):
+        DriverUpgradePolicySpec(maxUnavailable="nonsense")
+
+
+@pytest.mark.parametrize(
+    "value,total,round_up,expected",
+    [
+        ("25%", 8, True, 2),
+        ("25%", 10, True, 3),  # 2.5 rounds up
+        ("25%", 10, False, 2),
+        ("100%", 7, True, 7),
+        ("0%", 5, True, 0),
+        (3, 100, True, 3),
+        (None, 100, True, 0),
+    ],
+)
+def test_int_or_percent_scaling(value, total, round_up, expected):
+    assert IntOrString.scaled_value(value, total, round_up) == expected
+
+
+def test_deep_copy_is_independent():
+    p = DriverUpgradePolicySpec(drainSpec={"enable": True})
+    q = p.deep_copy()
+    q.drain_spec.enable = False
+    assert p.drain_spec.enable is True

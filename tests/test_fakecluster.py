@@ -1,0 +1,220 @@
+"""Tests for the in-memory apiserver substrate (core.fakecluster) and the
+selector/patch helpers (core.meta)."""
+
+import pytest
+
+from k8s_operator_libs_amd.core import (
+    AlreadyExistsError,
+    ConflictError,
+    FakeClient,
+    FakeCluster,
+    NotFoundError,
+)
+from k8s_operator_libs_amd.core.meta import (
+    FieldSelector,
+    LabelSelector,
+    json_merge_patch,
+)
+
+
+def mk_node(name, labels=None, unschedulable=False):
+    n = {
+        "apiVersion": "v1",
+        "kind": "Node",
+        "metadata": {"name": name, "labels": labels or {}},
+        "spec": {},
+        "status": {"conditions": [{"type": "Ready", "status": "True"}]},
+    }
+    if unschedulable:
+        n["spec"]["unschedulable"] = True
+    return n
+
+
+def mk_pod(name, node="", namespace="default", labels=None, phase="Running"):
+    return {
+        "apiVersion": "v1",
+        "kind": "Pod",
+        "metadata": {"name": name, "namespace": namespace, "labels": labels or {}},
+        "spec": {"nodeName": node},
+        "status": {"phase": phase},
+    }
+
+
+class TestCrud:
+    def test_create_get_roundtrip(self, cluster):
+        created = cluster.create(mk_node("n1", {"gpu": "mi355x"}))
+        assert created["metadata"]["uid"]
+        assert created["metadata"]["resourceVersion"]
+        got = cluster.get("v1", "Node", "n1")
+        assert got["metadata"]["labels"]["gpu"] == "mi355x"
+
+    def test_create_duplicate_fails(self, cluster):
+        cluster.create(mk_node("n1"))
+        with pytest.raises(AlreadyExistsError):
+            cluster.create(mk_node("n1"))
+
+    def test_get_missing_raises(self, cluster):
+        with pytest.raises(NotFoundError):
+            cluster.get("v1", "Node", "nope")
+
+    def test_generate_name(self, cluster):
+        p = cluster.create(
+            {"apiVersion": "v1", "kind": "Pod",
+             "metadata": {"generateName": "driver-", "namespace": "ops"},
+             "spec": {}}
+        )
+        assert p["metadata"]["name"].startswith("driver-")
+
+    def test_update_bumps_rv_and_conflicts_on_stale(self, cluster):
+        cluster.create(mk_node("n1"))
+        a = cluster.get("v1", "Node", "n1")
+        b = cluster.get("v1", "Node", "n1")
+        a["metadata"]["labels"]["x"] = "1"
+        cluster.update(a)
+        b["metadata"]["labels"]["x"] = "2"
+        with pytest.raises(ConflictError):
+            cluster.update(b)
+
+    def test_delete(self, cluster):
+        cluster.create(mk_node("n1"))
+        cluster.delete("v1", "Node", "n1")
+        with pytest.raises(NotFoundError):
+            cluster.get("v1", "Node", "n1")
+        with pytest.raises(NotFoundError):
+            cluster.delete("v1", "Node", "n1")
+
+    def test_namespacing(self, cluster):
+        cluster.create(mk_pod("p", namespace="a"))
+        cluster.create(mk_pod("p", namespace="b"))
+        assert len(cluster.list("v1", "Pod")) == 2
+        assert len(cluster.list("v1", "Pod", namespace="a")) == 1
+
+
+class TestListSelectors:
+    def test_label_selector(self, cluster):
+        cluster.create(mk_pod("p1", labels={"app": "train", "tier": "gpu"}))
+        cluster.create(mk_pod("p2", labels={"app": "serve"}))
+        out = cluster.list("v1", "Pod", label_selector="app=train")
+        assert [p["metadata"]["name"] for p in out] == ["p1"]
+        out = cluster.list("v1", "Pod", label_selector="app in (train,serve)")
+        assert len(out) == 2
+        out = cluster.list("v1", "Pod", label_selector="tier")
+        assert [p["metadata"]["name"] for p in out] == ["p1"]
+
+    def test_field_selector_node_name(self, cluster):
+        cluster.create(mk_pod("p1", node="n1"))
+        cluster.create(mk_pod("p2", node="n2"))
+        out = cluster.list("v1", "Pod", field_selector="spec.nodeName=n1")
+        assert [p["metadata"]["name"] for p in out] == ["p1"]
+
+
+class TestPatch:
+    def test_merge_patch_labels(self, cluster):
+        cluster.create(mk_node("n1", {"a": "1", "b": "2"}))
+        cluster.patch("v1", "Node", "n1", {"metadata": {"labels": {"a": "9", "c": "3"}}})
+        got = cluster.get("v1", "Node", "n1")
+        assert got["metadata"]["labels"] == {"a": "9", "b": "2", "c": "3"}
+
+    def test_merge_patch_null_deletes(self, cluster):
+        cluster.create(mk_node("n1", {"a": "1"}))
+        cluster.patch("v1", "Node", "n1", {"metadata": {"labels": {"a": None}}})
+        assert "a" not in cluster.get("v1", "Node", "n1")["metadata"]["labels"]
+
+    def test_optimistic_lock_patch(self, cluster):
+        cluster.create(mk_node("n1"))
+        rv = cluster.get("v1", "Node", "n1")["metadata"]["resourceVersion"]
+        cluster.patch("v1", "Node", "n1", {"metadata": {"resourceVersion": rv, "labels": {"x": "1"}}})
+        with pytest.raises(ConflictError):
+            cluster.patch(
+                "v1", "Node", "n1",
+                {"metadata": {"resourceVersion": rv, "labels": {"x": "2"}}},
+            )
+
+
+class TestFinalizers:
+    def test_delete_with_finalizer_defers(self, cluster):
+        nm = {
+            "apiVersion": "maintenance.amd.com/v1alpha1",
+            "kind": "NodeMaintenance",
+            "metadata": {"name": "m1", "namespace": "ops", "finalizers": ["ext/guard"]},
+            "spec": {"nodeName": "n1"},
+        }
+        cluster.create(nm)
+        cluster.delete("maintenance.amd.com/v1alpha1", "NodeMaintenance", "m1", "ops")
+        got = cluster.get("maintenance.amd.com/v1alpha1", "NodeMaintenance", "m1", "ops")
+        assert "deletionTimestamp" in got["metadata"]
+        # external operator removes the finalizer -> object goes away
+        cluster.patch(
+            "maintenance.amd.com/v1alpha1", "NodeMaintenance", "m1",
+            {"metadata": {"finalizers": []}}, "ops",
+        )
+        with pytest.raises(NotFoundError):
+            cluster.get("maintenance.amd.com/v1alpha1", "NodeMaintenance", "m1", "ops")
+
+
+class TestWatchAndEviction:
+    def test_watch_sees_lifecycle(self, cluster):
+        w = cluster.watch("v1", "Node")
+        cluster.create(mk_node("n1"))
+        cluster.patch("v1", "Node", "n1", {"metadata": {"labels": {"s": "1"}}})
+        cluster.delete("v1", "Node", "n1")
+        events = [w.next(timeout=1)[0] for _ in range(3)]
+        assert events == ["ADDED", "MODIFIED", "DELETED"]
+        w.stop()
+
+    def test_evict_pod(self, cluster):
+        cluster.create(mk_pod("p1"))
+        cluster.evict_pod("p1", "default")
+        with pytest.raises(NotFoundError):
+            cluster.get("v1", "Pod", "p1", "default")
+
+
+class TestCrdRegistration:
+    def test_created_crd_becomes_servable(self, cluster):
+        crd = {
+            "apiVersion": "apiextensions.k8s.io/v1",
+            "kind": "CustomResourceDefinition",
+            "metadata": {"name": "widgets.amd.com"},
+            "spec": {
+                "group": "amd.com",
+                "scope": "Namespaced",
+                "names": {"kind": "Widget", "plural": "widgets"},
+                "versions": [{"name": "v1", "served": True}],
+            },
+        }
+        cluster.create(crd)
+        assert cluster.lookup_by_plural("amd.com/v1", "widgets") == "Widget"
+        cluster.create(
+            {"apiVersion": "amd.com/v1", "kind": "Widget",
+             "metadata": {"name": "w1", "namespace": "default"}}
+        )
+        assert cluster.get("amd.com/v1", "Widget", "w1", "default")
+
+
+class TestClientFacade:
+    def test_typed_helpers(self, client):
+        client.create(mk_node("n1"))
+        client.create(mk_pod("p1", node="n1"))
+        assert client.get_node("n1")["metadata"]["name"] == "n1"
+        assert len(client.list_pods(field_selector="spec.nodeName=n1")) == 1
+
+
+class TestMetaHelpers:
+    def test_label_selector_negations(self):
+        sel = LabelSelector("a!=x,!b,c")
+        assert sel.matches({"c": "1"})
+        assert sel.matches({"a": "y", "c": "1"})
+        assert not sel.matches({"a": "x", "c": "1"})
+        assert not sel.matches({"b": "1", "c": "1"})
+        assert not sel.matches({})
+
+    def test_field_selector(self):
+        fs = FieldSelector("spec.nodeName=n1,status.phase!=Failed")
+        assert fs.matches_object({"spec": {"nodeName": "n1"}, "status": {"phase": "Running"}})
+        assert not fs.matches_object({"spec": {"nodeName": "n2"}, "status": {"phase": "Running"}})
+        assert not fs.matches_object({"spec": {"nodeName": "n1"}, "status": {"phase": "Failed"}})
+
+    def test_json_merge_patch_nested(self):
+        t = {"a": {"b": 1, "c": 2}, "d": 3}
+        json_merge_patch(t, {"a": {"b": None, "e": 4}, "d": 5})
+        assert t == {"a": {"c": 2, "e": 4}, "d": 5}
