@@ -1,0 +1,107 @@
+"""Async node drain manager.
+
+Capability parity with the reference's ``pkg/upgrade/drain_manager.go``: per
+node not already draining, a worker thread cordons the node, runs the drain
+engine (:mod:`k8s_operator_libs_amd.upgrade.drain`), and moves the node to
+pod-restart-required on success or upgrade-failed on error
+(drain_manager.go:98-137).  The in-progress StringSet deduplicates across
+reconcile ticks while a drain is still running.
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+from ..api.upgrade.v1alpha1 import DrainSpec
+from ..core import meta
+from ..core.client import Client
+from ..core.events import EVENT_TYPE_NORMAL, EVENT_TYPE_WARNING, log_event, log_eventf
+from ..core.meta import K8sObject
+from . import consts, util
+from .drain import drain_node
+from .node_state_provider import NodeUpgradeStateProvider
+
+logger = logging.getLogger(__name__)
+
+
+@dataclass
+class DrainConfiguration:
+    """(drain_manager.go:33-36)"""
+
+    spec: Optional[DrainSpec] = None
+    nodes: List[K8sObject] = field(default_factory=list)
+
+
+class DrainManager:
+    def __init__(
+        self,
+        client: Client,
+        node_state_provider: NodeUpgradeStateProvider,
+        event_recorder: Optional[object] = None,
+    ) -> None:
+        self._client = client
+        self._provider = node_state_provider
+        self._recorder = event_recorder
+        self._draining_nodes = util.StringSet()
+        self._workers: List[threading.Thread] = []
+        self._workers_lock = threading.Lock()
+
+    def schedule_nodes_drain(self, config: DrainConfiguration) -> None:
+        if config.spec is None:
+            raise ValueError("drain spec should not be nil")
+        if not config.spec.enable:
+            logger.info("drain is disabled; nothing to schedule")
+            return
+        if not config.nodes:
+            logger.info("no nodes scheduled for drain")
+            return
+        for node in config.nodes:
+            node_name = meta.name(node)
+            if not self._draining_nodes.add_if_absent(node_name):
+                logger.info("node %s is already draining, skipping", node_name)
+                continue
+            t = threading.Thread(
+                target=self._drain_worker, args=(node, config.spec), daemon=True
+            )
+            with self._workers_lock:
+                self._workers = [w for w in self._workers if w.is_alive()]
+                self._workers.append(t)
+            t.start()
+
+    def _drain_worker(self, node: K8sObject, spec: DrainSpec) -> None:
+        node_name = meta.name(node)
+        try:
+            try:
+                # Cordon first: drains only make sense on unschedulable nodes
+                # (drain_manager.go:109-116 runs RunCordonOrUncordon first).
+                self._client.patch("v1", "Node", node_name, {"spec": {"unschedulable": True}})
+                node.setdefault("spec", {})["unschedulable"] = True
+                log_event(self._recorder, node, EVENT_TYPE_NORMAL, util.get_event_reason(),
+                          "Node drain started for driver upgrade")
+                drain_node(self._client, node_name, spec)
+            except Exception as exc:
+                logger.error("drain failed for node %s: %s", node_name, exc)
+                log_eventf(self._recorder, node, EVENT_TYPE_WARNING, util.get_event_reason(),
+                           "Node drain failed: {}", exc)
+                self._provider.change_node_upgrade_state(node, consts.UPGRADE_STATE_FAILED)
+                return
+            self._provider.change_node_upgrade_state(
+                node, consts.UPGRADE_STATE_POD_RESTART_REQUIRED
+            )
+            log_event(self._recorder, node, EVENT_TYPE_NORMAL, util.get_event_reason(),
+                      "Node drain completed for driver upgrade")
+        except Exception:
+            logger.exception("drain worker crashed for node %s", node_name)
+        finally:
+            self._draining_nodes.remove(node_name)
+
+    def wait_idle(self, timeout: float = 60.0) -> None:
+        deadline = time.monotonic() + timeout
+        with self._workers_lock:
+            workers = list(self._workers)
+        for w in workers:
+            w.join(max(0.0, deadline - time.monotonic()))
