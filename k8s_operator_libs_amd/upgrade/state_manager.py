@@ -1,0 +1,212 @@
+"""Cluster upgrade state manager: the top-level facade.
+
+Capability parity with the reference's ``pkg/upgrade/upgrade_state.go``:
+``build_state`` snapshots DaemonSets -> driver pods -> nodes into a
+:class:`~k8s_operator_libs_amd.upgrade.common_manager.ClusterUpgradeState`
+keyed by the state label, and ``apply_state`` runs the eleven Process* phases
+in fixed order, dispatching mode-specific phases to the in-place or requestor
+implementation.  Both calls are stateless and idempotent: a consumer
+operator's reconcile is simply ``apply_state(build_state(...), policy)``.
+"""
+
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass, field
+from typing import Dict, Optional
+
+from ..api.upgrade.v1alpha1 import DriverUpgradePolicySpec
+from ..core import meta
+from ..core.client import Client
+from ..core.meta import K8sObject
+from . import consts, util
+from .common_manager import (
+    ClusterUpgradeState,
+    CommonUpgradeManager,
+    NodeUpgradeState,
+    is_orphaned_pod,
+)
+from .inplace import InplaceNodeStateManager
+from .pod_manager import PodDeletionFilter, PodManager
+from .requestor import RequestorNodeStateManager, RequestorOptions
+from .validation_manager import ValidationManager
+
+logger = logging.getLogger(__name__)
+
+
+class BuildStateError(Exception):
+    pass
+
+
+@dataclass
+class StateOptions:
+    """(upgrade_state.go:94-96)"""
+
+    requestor: RequestorOptions = field(default_factory=RequestorOptions)
+
+
+class ClusterUpgradeStateManager:
+    """(ClusterUpgradeStateManagerImpl, upgrade_state.go:35-92)"""
+
+    def __init__(
+        self,
+        client: Client,
+        event_recorder: Optional[object] = None,
+        options: Optional[StateOptions] = None,
+    ) -> None:
+        self.opts = options or StateOptions()
+        self.common = CommonUpgradeManager(client, event_recorder)
+        self.inplace = InplaceNodeStateManager(self.common)
+        self.requestor: Optional[RequestorNodeStateManager] = None
+        if self.opts.requestor.use_maintenance_operator:
+            self.requestor = RequestorNodeStateManager(self.common, self.opts.requestor)
+
+    # -- builder options (upgrade_state.go:329-350) --------------------------
+
+    def with_pod_deletion_enabled(self, filter_: PodDeletionFilter) -> "ClusterUpgradeStateManager":
+        if filter_ is None:
+            logger.warning("cannot enable PodDeletion state: filter is None")
+            return self
+        self.common.pod_manager = PodManager(
+            self.common.client, self.common.node_state_provider,
+            pod_deletion_filter=filter_, event_recorder=self.common.event_recorder,
+        )
+        self.common.pod_deletion_state_enabled = True
+        return self
+
+    def with_validation_enabled(self, pod_selector: str) -> "ClusterUpgradeStateManager":
+        if not pod_selector:
+            logger.warning("cannot enable Validation state: podSelector is empty")
+            return self
+        self.common.validation_manager = ValidationManager(
+            self.common.client, self.common.node_state_provider,
+            pod_selector=pod_selector, event_recorder=self.common.event_recorder,
+        )
+        self.common.validation_state_enabled = True
+        return self
+
+    # -- snapshot construction (upgrade_state.go:99-164) ----------------------
+
+    def build_state(self, namespace: str, driver_labels: Dict[str, str]) -> ClusterUpgradeState:
+        """One GET-free snapshot: DaemonSets and driver pods are listed, then
+        each pod's node is fetched.  Fails if any driver DaemonSet has
+        unscheduled pods (upgrade_state.go:128-131)."""
+        state = ClusterUpgradeState()
+        daemonsets = self.common.get_driver_daemonsets(namespace, driver_labels)
+        selector = ",".join(f"{k}={v}" for k, v in sorted(driver_labels.items()))
+        pods = self.common.client.list_pods(namespace=namespace, label_selector=selector)
+
+        filtered_pods = []
+        for ds in daemonsets.values():
+            ds_pods = self.common.get_pods_owned_by_ds(ds, pods)
+            desired = ds.get("status", {}).get("desiredNumberScheduled", 0)
+            if desired != len(ds_pods):
+                raise BuildStateError(
+                    f"driver DaemonSet {meta.name(ds)} should not have unscheduled pods "
+                    f"(desired {desired}, found {len(ds_pods)})"
+                )
+            filtered_pods.extend(ds_pods)
+        filtered_pods.extend(self.common.get_orphaned_pods(pods))
+
+        state_label = util.get_upgrade_state_label_key()
+        # Node fetches are deduplicated: one GET per node even if several
+        # driver pods (multiple driver DaemonSets) land on it.
+        node_cache: Dict[str, K8sObject] = {}
+        for pod in filtered_pods:
+            ds = None
+            if not is_orphaned_pod(pod):
+                ds = daemonsets.get(meta.owner_references(pod)[0].get("uid"))
+            node_name = pod.get("spec", {}).get("nodeName", "")
+            if not node_name and pod.get("status", {}).get("phase") == "Pending":
+                logger.info("driver pod %s has no NodeName, skipping", meta.name(pod))
+                continue
+            node = node_cache.get(node_name)
+            if node is None:
+                node = self.common.node_state_provider.get_node(node_name)
+                node_cache[node_name] = node
+            node_state = NodeUpgradeState(node=node, driver_pod=pod, driver_daemonset=ds)
+            if self.requestor is not None:
+                node_state.node_maintenance = self.requestor.get_node_maintenance_obj(
+                    node_name
+                )
+            state.add(meta.get_label(node, state_label), node_state)
+        return state
+
+    # -- one state-machine tick (upgrade_state.go:171-281) --------------------
+
+    def apply_state(
+        self,
+        current_state: Optional[ClusterUpgradeState],
+        upgrade_policy: Optional[DriverUpgradePolicySpec],
+    ) -> None:
+        if current_state is None:
+            raise ValueError("currentState should not be empty")
+        if upgrade_policy is None or not upgrade_policy.auto_upgrade:
+            logger.info("driver auto upgrade is disabled, skipping")
+            return
+
+        logger.info(
+            "node states: %s",
+            {s or "Unknown": len(v) for s, v in current_state.node_states.items()},
+        )
+
+        common = self.common
+        common.process_done_or_unknown_nodes(current_state, consts.UPGRADE_STATE_UNKNOWN)
+        common.process_done_or_unknown_nodes(current_state, consts.UPGRADE_STATE_DONE)
+        self._process_upgrade_required_nodes_wrapper(current_state, upgrade_policy)
+        common.process_cordon_required_nodes(current_state)
+        common.process_wait_for_jobs_required_nodes(
+            current_state, upgrade_policy.wait_for_completion
+        )
+        drain_enabled = (
+            upgrade_policy.drain_spec is not None and upgrade_policy.drain_spec.enable
+        )
+        common.process_pod_deletion_required_nodes(
+            current_state, upgrade_policy.pod_deletion, drain_enabled
+        )
+        common.process_drain_nodes(current_state, upgrade_policy.drain_spec)
+        self._process_node_maintenance_required_nodes_wrapper(current_state)
+        common.process_pod_restart_nodes(current_state)
+        common.process_upgrade_failed_nodes(current_state)
+        common.process_validation_required_nodes(current_state)
+        self._process_uncordon_required_nodes_wrapper(current_state)
+
+    # -- mode dispatch (upgrade_state.go:287-325) -----------------------------
+
+    def _process_upgrade_required_nodes_wrapper(
+        self, state: ClusterUpgradeState, policy: DriverUpgradePolicySpec
+    ) -> None:
+        if self.requestor is not None:
+            self.requestor.process_upgrade_required_nodes(state, policy)
+        else:
+            self.inplace.process_upgrade_required_nodes(state, policy)
+
+    def _process_node_maintenance_required_nodes_wrapper(
+        self, state: ClusterUpgradeState
+    ) -> None:
+        if self.requestor is not None:
+            self.requestor.process_node_maintenance_required_nodes(state)
+
+    def _process_uncordon_required_nodes_wrapper(self, state: ClusterUpgradeState) -> None:
+        # Inplace ALWAYS runs first so nodes that began an in-place upgrade
+        # before requestor mode was enabled still complete
+        # (upgrade_state.go:311-325).
+        self.inplace.process_uncordon_required_nodes(state)
+        if self.requestor is not None:
+            self.requestor.process_uncordon_required_nodes(state)
+
+    # -- metrics convenience ---------------------------------------------------
+
+    def counts(self, state: ClusterUpgradeState) -> Dict[str, int]:
+        """Exported so consumers can publish them as metrics (the reference
+        exposes GetUpgrades{InProgress,Done,Failed,Pending} getters)."""
+        return {
+            "total": self.common.get_total_managed_nodes(state),
+            "in_progress": self.common.get_upgrades_in_progress(state),
+            "done": self.common.get_upgrades_done(state),
+            "failed": self.common.get_upgrades_failed(state),
+            "pending": self.common.get_upgrades_pending(state),
+        }
+
+    def wait_idle(self, timeout: float = 60.0) -> None:
+        self.common.wait_idle(timeout)

@@ -94,10 +94,10 @@ class PodBuilder:
         )
         return self
 
-    def with_owner_reference(self, kind, name, controller=True):
+    def with_owner_reference(self, kind, name, controller=True, uid=None):
         self.obj["metadata"].setdefault("ownerReferences", []).append(
             {"apiVersion": "apps/v1", "kind": kind, "name": name,
-             "uid": str(uuid.uuid4()), "controller": controller}
+             "uid": uid or str(uuid.uuid4()), "controller": controller}
         )
         return self
 
@@ -155,7 +155,8 @@ def driver_pod_for(ds, node, hash_="rev1", ready=True, namespace=None):
         PodBuilder(f"{ds['metadata']['name']}-{node}", node=node,
                    namespace=namespace or ds["metadata"]["namespace"])
         .with_labels(dict(ds["spec"]["selector"]["matchLabels"]))
-        .with_owner_reference("DaemonSet", ds["metadata"]["name"])
+        .with_owner_reference("DaemonSet", ds["metadata"]["name"],
+                              uid=ds["metadata"].get("uid"))
         .with_revision_hash(hash_)
     )
     if not ready:
