@@ -1,0 +1,357 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: 8-node rolling amdgpu-driver upgrade (BASELINE config #3).
+
+Measures the metric BASELINE.json names — rolling-upgrade wall-clock and
+reconcile p50 for an 8-node amdgpu driver bump with maxParallelUpgrades=2,
+cordon+drain of synthetic ``amd.com/gpu`` pods — on a synthetic cluster
+served by the in-memory apiserver (this library's envtest).  The reference
+publishes no numbers, so these runs establish the baseline.
+
+One *step* = one complete rolling upgrade of an 8-node cluster: every node
+travels upgrade-required -> cordon -> wait-for-jobs -> pod-deletion -> drain
+-> driver-pod-restart -> validate -> uncordon -> done, driven by repeated
+build_state/apply_state reconcile ticks.  With a GPU present, each node's
+validation step actually executes the native gfx950 health-check kernels
+(MFMA matrix-core smoke) on this rank's device, so the number is grounded in
+real MI355X validation work; without a GPU the validator pod is marked ready
+directly (CPU-only mode, reported in the JSON).
+
+Scaling is weak: each rank runs an independent 8-node cluster upgrade
+(one rank per GPU over torch.distributed, rendezvous provided by the driver).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import time
+
+from k8s_operator_libs_amd.api.upgrade.v1alpha1 import DriverUpgradePolicySpec
+from k8s_operator_libs_amd.core import FakeClient
+from k8s_operator_libs_amd.upgrade import consts, util
+from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
+from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
+
+DRIVER_NS = "amd-gpu-operator"
+DRIVER_LABELS = {"app": "amdgpu-driver-daemonset"}
+VALIDATOR_SELECTOR = "app=amd-gpu-validator"
+
+
+# ---------------------------------------------------------------------------
+# synthetic cluster construction
+# ---------------------------------------------------------------------------
+
+def _make_cluster(client, n_nodes, old_hash, new_hash, gpu_pods_per_node=2):
+    c = client.cluster
+    labels = dict(DRIVER_LABELS)
+    ds = c.create({
+        "apiVersion": "apps/v1", "kind": "DaemonSet",
+        "metadata": {"name": "amdgpu-driver", "namespace": DRIVER_NS,
+                     "labels": labels},
+        "spec": {"selector": {"matchLabels": labels},
+                 "template": {"metadata": {"labels": labels}}},
+        "status": {"desiredNumberScheduled": n_nodes},
+    })
+    for rev, h in ((1, old_hash), (2, new_hash)):
+        c.create({
+            "apiVersion": "apps/v1", "kind": "ControllerRevision",
+            "metadata": {"name": f"amdgpu-driver-{h}", "namespace": DRIVER_NS,
+                         "labels": dict(labels)},
+            "revision": rev,
+        })
+    for i in range(n_nodes):
+        node = f"mi355x-{i}"
+        c.create({
+            "apiVersion": "v1", "kind": "Node",
+            "metadata": {"name": node, "labels": {}, "annotations": {}},
+            "spec": {},
+            "status": {"conditions": [{"type": "Ready", "status": "True"}],
+                       "capacity": {"amd.com/gpu": "8"}},
+        })
+        pod_labels = dict(labels)
+        pod_labels["controller-revision-hash"] = old_hash
+        c.create({
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": f"amdgpu-driver-{node}", "namespace": DRIVER_NS,
+                         "labels": pod_labels,
+                         "ownerReferences": [{"apiVersion": "apps/v1",
+                                              "kind": "DaemonSet",
+                                              "name": "amdgpu-driver",
+                                              "uid": ds["metadata"]["uid"],
+                                              "controller": True}]},
+            "spec": {"nodeName": node,
+                     "containers": [{"name": "driver", "image": "amdgpu-dkms:old"}]},
+            "status": {"phase": "Running",
+                       "containerStatuses": [{"name": "driver", "ready": True,
+                                              "restartCount": 0}]},
+        })
+        # synthetic GPU workload pods that must be evicted before the bump
+        for j in range(gpu_pods_per_node):
+            c.create({
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": f"train-{node}-{j}", "namespace": "default",
+                             "labels": {"app": "training"},
+                             "ownerReferences": [{"apiVersion": "apps/v1",
+                                                  "kind": "ReplicaSet",
+                                                  "name": "train-rs",
+                                                  "uid": f"rs-uid-{i}-{j}",
+                                                  "controller": True}]},
+                "spec": {"nodeName": node,
+                         "containers": [{"name": "t", "image": "train",
+                                         "resources": {"limits": {"amd.com/gpu": "8"}}}]},
+                "status": {"phase": "Running",
+                           "containerStatuses": [{"name": "t", "ready": True,
+                                                  "restartCount": 0}]},
+            })
+        # validation pod (not ready until the GPU health check passes)
+        c.create({
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": f"validator-{node}", "namespace": DRIVER_NS,
+                         "labels": {"app": "amd-gpu-validator"}},
+            "spec": {"nodeName": node,
+                     "containers": [{"name": "v", "image": "amd-gpu-validator"}]},
+            "status": {"phase": "Running",
+                       "containerStatuses": [{"name": "v", "ready": False,
+                                              "restartCount": 0}]},
+        })
+    return ds
+
+
+class _DsController:
+    """Recreates deleted driver pods with the new revision (the DaemonSet
+    controller + kubelet role, like envtest tests do by hand)."""
+
+    def __init__(self, cluster, ds, new_hash):
+        import threading
+
+        self.cluster, self.ds, self.new_hash = cluster, ds, new_hash
+        self._lock = threading.Lock()
+        cluster.add_change_hook(self._on_change)
+
+    def _on_change(self, event_type, obj):
+        if event_type != "DELETED" or obj.get("kind") != "Pod":
+            return
+        refs = obj.get("metadata", {}).get("ownerReferences") or []
+        if not refs or refs[0].get("uid") != self.ds["metadata"]["uid"]:
+            return
+        with self._lock:
+            labels = dict(DRIVER_LABELS)
+            labels["controller-revision-hash"] = self.new_hash
+            self.cluster.create({
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": obj["metadata"]["name"],
+                             "namespace": obj["metadata"]["namespace"],
+                             "labels": labels,
+                             "ownerReferences": refs},
+                "spec": {"nodeName": obj["spec"].get("nodeName", ""),
+                         "containers": [{"name": "driver", "image": "amdgpu-dkms:new"}]},
+                "status": {"phase": "Running",
+                           "containerStatuses": [{"name": "driver", "ready": True,
+                                                  "restartCount": 0}]},
+            })
+
+
+def _run_gpu_validation(device):
+    """The validation-pod payload: native gfx950 MFMA smoke on this rank's
+    GPU.  Raises if the native extension is missing on a GPU box."""
+    from k8s_operator_libs_amd.validation import load_native_validator
+
+    native = load_native_validator()
+    if native is None:
+        raise RuntimeError("GPU validation requested but native validator missing")
+    err = native.mfma_f32_check(device)
+    if err > 1e-6:
+        raise RuntimeError(f"MFMA validation failed: err={err}")
+    return err
+
+
+def run_rolling_upgrade_benchmark(
+    n_nodes=8, steps=10, warmup=2, max_parallel=2, gpu_validate=False,
+    device=0, gpu_pods_per_node=2, print_json=True,
+):
+    """Run `warmup` untimed + `steps` timed full rolling upgrades; returns a
+    result dict (single-process path; bench main() adds distribution)."""
+    policy = DriverUpgradePolicySpec.model_validate({
+        "autoUpgrade": True,
+        "maxParallelUpgrades": max_parallel,
+        "maxUnavailable": "50%",
+        "podDeletion": {"force": False, "deleteEmptyDir": True},
+        "drainSpec": {"enable": True, "timeoutSeconds": 300},
+    })
+    state_key = util.get_upgrade_state_label_key()
+
+    wall_times = []
+    tick_times = []
+    ticks_per_upgrade = []
+    completed = 0
+
+    for it in range(warmup + steps):
+        timed = it >= warmup
+        client = FakeClient()
+        ds = _make_cluster(client, n_nodes, "oldrev", "newrev", gpu_pods_per_node)
+        _DsController(client.cluster, ds, "newrev")
+        manager = (
+            ClusterUpgradeStateManager(client)
+            .with_pod_deletion_enabled(gpu_pod_deletion_filter)
+            .with_validation_enabled(VALIDATOR_SELECTOR)
+        )
+        t0 = time.perf_counter()
+        ticks = 0
+        while True:
+            tick_start = time.perf_counter()
+            state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+            manager.apply_state(state, policy)
+            manager.wait_idle()
+            if timed:
+                tick_times.append(time.perf_counter() - tick_start)
+            ticks += 1
+            # validation pods: run the real GPU check, then mark Ready
+            for ns_ in state.nodes_in(consts.UPGRADE_STATE_VALIDATION_REQUIRED):
+                node_name = ns_.node["metadata"]["name"]
+                if gpu_validate:
+                    _run_gpu_validation(device)
+                client.cluster.patch(
+                    "v1", "Pod", f"validator-{node_name}",
+                    {"status": {"containerStatuses": [
+                        {"name": "v", "ready": True, "restartCount": 0}]}},
+                    DRIVER_NS,
+                )
+            done = sum(
+                1 for n in client.list_nodes()
+                if n["metadata"]["labels"].get(state_key) == consts.UPGRADE_STATE_DONE
+            )
+            if done == n_nodes:
+                break
+            if ticks > 50 * n_nodes:
+                raise RuntimeError(f"upgrade did not converge after {ticks} ticks")
+        elapsed = time.perf_counter() - t0
+        if timed:
+            wall_times.append(elapsed)
+            ticks_per_upgrade.append(ticks)
+            completed += 1
+
+    return {
+        "upgrades_completed": completed,
+        "mean_wall_s": statistics.mean(wall_times) if wall_times else 0.0,
+        "wall_times": wall_times,
+        "reconcile_p50_ms": (
+            statistics.median(tick_times) * 1000 if tick_times else 0.0
+        ),
+        "reconcile_p99_ms": (
+            sorted(tick_times)[int(len(tick_times) * 0.99)] * 1000 if tick_times else 0.0
+        ),
+        "mean_ticks_per_upgrade": (
+            statistics.mean(ticks_per_upgrade) if ticks_per_upgrade else 0.0
+        ),
+    }
+
+
+# ---------------------------------------------------------------------------
+# distributed entry point (driver contract)
+# ---------------------------------------------------------------------------
+
+def main():
+    parser = argparse.ArgumentParser(description=__doc__)
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=10)
+    parser.add_argument("--warmup", type=int, default=2)
+    parser.add_argument("--nodes", type=int, default=8, help="simulated nodes per cluster")
+    parser.add_argument("--max-parallel", type=int, default=2)
+    parser.add_argument("--no-gpu-validate", action="store_true")
+    args = parser.parse_args()
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    have_cuda = torch.cuda.is_available()
+    gpu_validate = have_cuda and not args.no_gpu_validate
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if have_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+        if have_cuda:
+            torch.cuda.set_device(local_rank)
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if have_cuda:
+            torch.cuda.synchronize()
+
+    # warmup (untimed) + timed section bracketed by barrier+sync on both sides
+    run_rolling_upgrade_benchmark(
+        n_nodes=args.nodes, steps=0, warmup=args.warmup,
+        max_parallel=args.max_parallel, gpu_validate=gpu_validate,
+        device=local_rank if have_cuda else 0, print_json=False,
+    )
+    barrier_sync()
+    t0 = time.perf_counter()
+    result = run_rolling_upgrade_benchmark(
+        n_nodes=args.nodes, steps=args.steps, warmup=0,
+        max_parallel=args.max_parallel, gpu_validate=gpu_validate,
+        device=local_rank if have_cuda else 0, print_json=False,
+    )
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks for time-like metrics
+    if dist is not None:
+        t = torch.tensor([elapsed, result["mean_wall_s"], result["reconcile_p50_ms"]],
+                         dtype=torch.float64,
+                         device="cuda" if have_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed, mean_wall_s, reconcile_p50_ms = t.tolist()
+    else:
+        mean_wall_s = result["mean_wall_s"]
+        reconcile_p50_ms = result["reconcile_p50_ms"]
+
+    if rank == 0:
+        out = {
+            "metric": "8-node rolling-upgrade wall-clock",
+            "value": mean_wall_s,
+            "unit": "s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (in-memory apiserver, synthetic amd.com/gpu pods; "
+                    "reference publishes no baseline numbers)",
+            "config": {
+                "model": "amdgpu-driver-rolling-upgrade",
+                "nodes_per_cluster": args.nodes,
+                "maxParallelUpgrades": args.max_parallel,
+                "maxUnavailable": "50%",
+                "drain": True,
+                "podDeletion": True,
+                "gpu_pods_per_node": 2,
+                "gpu_validation": gpu_validate,
+                "global_batch": args.nodes * int(os.environ.get("WORLD_SIZE", "1")),
+                "parallelism": f"dp{world} (one independent simulated "
+                               f"{args.nodes}-node cluster per rank)",
+            },
+            "reconcile_p50_ms": reconcile_p50_ms,
+            "reconcile_p99_ms": result["reconcile_p99_ms"],
+            "mean_ticks_per_upgrade": result["mean_ticks_per_upgrade"],
+            "aggregate_node_upgrades_per_s": args.nodes * world / mean_wall_s
+            if mean_wall_s else None,
+        }
+        print(json.dumps(out))
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

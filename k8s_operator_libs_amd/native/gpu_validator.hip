@@ -1,0 +1,302 @@
+// gpu_validator.hip — native MI355X (gfx950) node health validator.
+//
+// AMD-native replacement for the NVML-based validation pods the reference
+// orchestrates ("validation pods run amd-smi/rocm-smi health
+// checks instead of NVML validators").  A validation pod on an MI355X node
+// runs these checks after a driver bump; the ValidationManager gates
+// uncordon on the pod reporting Ready.
+//
+// Checks:
+//   device_probe()        — device properties (arch, CUs, LDS, HBM)
+//   mfma_f32_check()      — v_mfma_f32_16x16x4_f32 tile vs exact CPU fmaf
+//                           chain (matrix cores, exact f32 numerics)
+//   mfma_bf16_check()     — v_mfma_f32_16x16x32_bf16 tile vs CPU reference
+//                           (the production-dtype matrix path)
+//   hbm_bandwidth_gbps()  — float4 streaming copy, grid sized to cover all
+//                           256 CUs across the 8 XCDs
+//   lds_roundtrip_check() — LDS store/load/barrier integrity
+//
+// Built standalone with hipcc (no torch linkage) via pybind11; the .so lives
+// in-tree so it travels to GPU nodes with the package.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cmath>
+#include <cstring>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string("HIP error at " #expr ": ") +       \
+                               hipGetErrorString(_e));                         \
+    }                                                                          \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// MFMA f32 smoke: one wave computes D = A·B + C for a 16x16 tile, K=4,
+// using v_mfma_f32_16x16x4_f32.  Lane layout (cdna_hip_programming.md §3):
+//   A operand: lane l supplies A[l&15][l>>4]        (one f32)
+//   B operand: lane l supplies B[l>>4][l&15]        (one f32)
+//   C/D:       lane l, reg r -> row=(l>>4)*4+r, col=l&15
+// ---------------------------------------------------------------------------
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__global__ void mfma_f32_16x16x4_kernel(const float* __restrict__ A,
+                                        const float* __restrict__ B,
+                                        float* __restrict__ D) {
+#if defined(__gfx950__)
+  int lane = threadIdx.x;   // one wave of 64
+  float a = A[(lane & 15) * 4 + (lane >> 4)];   // A is 16x4 row-major
+  float b = B[(lane >> 4) * 16 + (lane & 15)];  // B is 4x16 row-major
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  for (int r = 0; r < 4; ++r) {
+    int row = (lane >> 4) * 4 + r;
+    int col = lane & 15;
+    D[row * 16 + col] = c[r];
+  }
+#endif
+}
+
+// ---------------------------------------------------------------------------
+// MFMA bf16 smoke: v_mfma_f32_16x16x32_bf16, one wave, D = A·B.
+// Per cdna_hip_programming.md §3 each lane holds 8 bf16 of A and B (4 VGPRs)
+// and 4 f32 of C/D; C/D layout is dtype-independent (row=(l>>4)*4+r,
+// col=l&15).  A (16x32): lane l supplies A[l&15][(l>>4)*8 + i];
+// B (32x16): lane l supplies B[(l>>4)*8 + i][l&15].
+// ---------------------------------------------------------------------------
+
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+
+__global__ void mfma_bf16_16x16x32_kernel(const __bf16* __restrict__ A,
+                                          const __bf16* __restrict__ B,
+                                          float* __restrict__ D) {
+#if defined(__gfx950__)
+  int lane = threadIdx.x;
+  bf16x8 a, b;
+  for (int i = 0; i < 8; ++i) {
+    int k = (lane >> 4) * 8 + i;
+    a[i] = A[(lane & 15) * 32 + k];   // A is 16x32 row-major
+    b[i] = B[k * 16 + (lane & 15)];   // B is 32x16 row-major
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  for (int r = 0; r < 4; ++r) {
+    int row = (lane >> 4) * 4 + r;
+    int col = lane & 15;
+    D[row * 16 + col] = c[r];
+  }
+#endif
+}
+
+// ---------------------------------------------------------------------------
+// HBM streaming-copy bandwidth (float4 loads/stores, grid-stride loop).
+// Launch with blocks >> 256 so all 8 XCDs' CUs participate.
+// ---------------------------------------------------------------------------
+
+__global__ void bw_copy_kernel(const float4* __restrict__ src,
+                               float4* __restrict__ dst, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    dst[i] = src[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LDS integrity: stage a block through LDS with a permutation and read back.
+// ---------------------------------------------------------------------------
+
+__global__ void lds_roundtrip_kernel(const float* __restrict__ in,
+                                     float* __restrict__ out, int n) {
+  __shared__ float lds[1024];
+  int tid = threadIdx.x;
+  int base = blockIdx.x * 1024;
+  for (int i = tid; i < 1024; i += blockDim.x) {
+    lds[(i * 5 + 7) & 1023] = in[base + i] * 2.0f;
+  }
+  __syncthreads();
+  for (int i = tid; i < 1024; i += blockDim.x) {
+    out[base + i] = lds[(i * 5 + 7) & 1023];
+  }
+  (void)n;
+}
+
+// ---------------------------------------------------------------------------
+// Host-side checks
+// ---------------------------------------------------------------------------
+
+static py::dict device_probe(int device) {
+  hipDeviceProp_t prop;
+  HIP_CHECK(hipSetDevice(device));
+  HIP_CHECK(hipGetDeviceProperties(&prop, device));
+  py::dict d;
+  d["name"] = std::string(prop.name);
+  d["gcn_arch"] = std::string(prop.gcnArchName);
+  d["compute_units"] = prop.multiProcessorCount;
+  d["lds_per_block_kb"] = (int)(prop.sharedMemPerBlock / 1024);
+  d["hbm_total_gb"] = (double)prop.totalGlobalMem / (1024.0 * 1024.0 * 1024.0);
+  d["clock_mhz"] = prop.clockRate / 1000;
+  d["warp_size"] = prop.warpSize;
+  int count = 0;
+  HIP_CHECK(hipGetDeviceCount(&count));
+  d["device_count"] = count;
+  return d;
+}
+
+static double mfma_f32_check(int device) {
+  HIP_CHECK(hipSetDevice(device));
+  const int M = 16, N = 16, K = 4;
+  std::vector<float> hA(M * K), hB(K * N), hD(M * N), ref(M * N);
+  // asymmetric operands so transposed writes can't pass (guide §3 note)
+  for (int i = 0; i < M * K; ++i) hA[i] = 0.01f * (float)(i % 37) - 0.15f;
+  for (int i = 0; i < K * N; ++i) hB[i] = 0.02f * (float)((i * 7) % 23) - 0.2f;
+  // exact CPU reference: k-ordered fmaf chain (guide: bitwise equal)
+  for (int m = 0; m < M; ++m)
+    for (int n = 0; n < N; ++n) {
+      float acc = 0.f;
+      for (int k = 0; k < K; ++k) acc = fmaf(hA[m * K + k], hB[k * N + n], acc);
+      ref[m * N + n] = acc;
+    }
+  float *dA, *dB, *dD;
+  HIP_CHECK(hipMalloc(&dA, sizeof(float) * M * K));
+  HIP_CHECK(hipMalloc(&dB, sizeof(float) * K * N));
+  HIP_CHECK(hipMalloc(&dD, sizeof(float) * M * N));
+  HIP_CHECK(hipMemcpy(dA, hA.data(), sizeof(float) * M * K, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(dB, hB.data(), sizeof(float) * K * N, hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(mfma_f32_16x16x4_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dD);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipMemcpy(hD.data(), dD, sizeof(float) * M * N, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(dA));
+  HIP_CHECK(hipFree(dB));
+  HIP_CHECK(hipFree(dD));
+  double max_err = 0.0;
+  for (int i = 0; i < M * N; ++i)
+    max_err = std::max(max_err, (double)std::fabs(hD[i] - ref[i]));
+  return max_err;
+}
+
+static double mfma_bf16_check(int device) {
+  HIP_CHECK(hipSetDevice(device));
+  const int M = 16, N = 16, K = 32;
+  std::vector<float> hAf(M * K), hBf(K * N), hD(M * N), ref(M * N);
+  std::vector<uint16_t> hA(M * K), hB(K * N);
+  auto to_bf16 = [](float f) -> uint16_t {
+    uint32_t u;
+    std::memcpy(&u, &f, 4);
+    // round-to-nearest-even truncation
+    uint32_t rounded = u + 0x7FFF + ((u >> 16) & 1);
+    return (uint16_t)(rounded >> 16);
+  };
+  auto from_bf16 = [](uint16_t v) -> float {
+    uint32_t u = (uint32_t)v << 16;
+    float f;
+    std::memcpy(&f, &u, 4);
+    return f;
+  };
+  for (int i = 0; i < M * K; ++i) hAf[i] = 0.03f * (float)((i * 3) % 29) - 0.4f;
+  for (int i = 0; i < K * N; ++i) hBf[i] = 0.015f * (float)((i * 11) % 31) - 0.2f;
+  for (int i = 0; i < M * K; ++i) hA[i] = to_bf16(hAf[i]);
+  for (int i = 0; i < K * N; ++i) hB[i] = to_bf16(hBf[i]);
+  // f32 CPU reference over the bf16-quantized operands
+  for (int m = 0; m < M; ++m)
+    for (int n = 0; n < N; ++n) {
+      float acc = 0.f;
+      for (int k = 0; k < K; ++k)
+        acc = fmaf(from_bf16(hA[m * K + k]), from_bf16(hB[k * N + n]), acc);
+      ref[m * N + n] = acc;
+    }
+  uint16_t *dA, *dB;
+  float* dD;
+  HIP_CHECK(hipMalloc(&dA, sizeof(uint16_t) * M * K));
+  HIP_CHECK(hipMalloc(&dB, sizeof(uint16_t) * K * N));
+  HIP_CHECK(hipMalloc(&dD, sizeof(float) * M * N));
+  HIP_CHECK(hipMemcpy(dA, hA.data(), sizeof(uint16_t) * M * K, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(dB, hB.data(), sizeof(uint16_t) * K * N, hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(mfma_bf16_16x16x32_kernel, dim3(1), dim3(64), 0, 0,
+                     (const __bf16*)dA, (const __bf16*)dB, dD);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipMemcpy(hD.data(), dD, sizeof(float) * M * N, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(dA));
+  HIP_CHECK(hipFree(dB));
+  HIP_CHECK(hipFree(dD));
+  double max_err = 0.0;
+  for (int i = 0; i < M * N; ++i)
+    max_err = std::max(max_err, (double)std::fabs(hD[i] - ref[i]));
+  return max_err;
+}
+
+static double hbm_bandwidth_gbps(int device, double buf_mib, int iters) {
+  HIP_CHECK(hipSetDevice(device));
+  size_t bytes = (size_t)(buf_mib * 1024.0 * 1024.0);
+  size_t n = bytes / sizeof(float4);
+  bytes = n * sizeof(float4);
+  float4 *src, *dst;
+  HIP_CHECK(hipMalloc(&src, bytes));
+  HIP_CHECK(hipMalloc(&dst, bytes));
+  HIP_CHECK(hipMemset(src, 1, bytes));
+  int block = 256;
+  // >> 256 workgroups so every CU on all 8 XCDs gets work
+  int grid = 8192;
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  // warmup
+  hipLaunchKernelGGL(bw_copy_kernel, dim3(grid), dim3(block), 0, 0, src, dst, n);
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i)
+    hipLaunchKernelGGL(bw_copy_kernel, dim3(grid), dim3(block), 0, 0, src, dst, n);
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  HIP_CHECK(hipFree(src));
+  HIP_CHECK(hipFree(dst));
+  // read + write
+  double gb = 2.0 * (double)bytes * iters / 1e9;
+  return gb / ((double)ms / 1e3);
+}
+
+static bool lds_roundtrip_check(int device) {
+  HIP_CHECK(hipSetDevice(device));
+  const int blocks = 512, n = blocks * 1024;
+  std::vector<float> hin(n), hout(n);
+  for (int i = 0; i < n; ++i) hin[i] = (float)(i % 977) * 0.5f;
+  float *din, *dout;
+  HIP_CHECK(hipMalloc(&din, sizeof(float) * n));
+  HIP_CHECK(hipMalloc(&dout, sizeof(float) * n));
+  HIP_CHECK(hipMemcpy(din, hin.data(), sizeof(float) * n, hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(lds_roundtrip_kernel, dim3(blocks), dim3(256), 0, 0, din, dout, n);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipMemcpy(hout.data(), dout, sizeof(float) * n, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(din));
+  HIP_CHECK(hipFree(dout));
+  for (int i = 0; i < n; ++i)
+    if (hout[i] != hin[i] * 2.0f) return false;
+  return true;
+}
+
+PYBIND11_MODULE(_gpu_validator, m) {
+  m.doc() = "MI355X (gfx950) native GPU health validator";
+  m.def("device_probe", &device_probe, py::arg("device") = 0);
+  m.def("mfma_f32_check", &mfma_f32_check, py::arg("device") = 0,
+        "Max abs error of a v_mfma_f32_16x16x4_f32 tile vs exact CPU fmaf chain");
+  m.def("mfma_bf16_check", &mfma_bf16_check, py::arg("device") = 0,
+        "Max abs error of a v_mfma_f32_16x16x32_bf16 tile vs f32 CPU reference");
+  m.def("hbm_bandwidth_gbps", &hbm_bandwidth_gbps, py::arg("device") = 0,
+        py::arg("buf_mib") = 1024.0, py::arg("iters") = 10,
+        "Streaming float4 copy bandwidth in GB/s (read+write)");
+  m.def("lds_roundtrip_check", &lds_roundtrip_check, py::arg("device") = 0);
+}

@@ -1,0 +1,37 @@
+"""Benchmark harness smoke tests (CPU-only path)."""
+
+import json
+import subprocess
+import sys
+import os
+
+
+def test_rolling_upgrade_benchmark_function():
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    import bench
+
+    result = bench.run_rolling_upgrade_benchmark(
+        n_nodes=2, steps=1, warmup=0, max_parallel=1, gpu_validate=False,
+        print_json=False,
+    )
+    assert result["upgrades_completed"] == 1
+    assert result["mean_wall_s"] > 0
+    assert result["reconcile_p50_ms"] > 0
+
+
+def test_bench_cli_prints_contract_json():
+    repo = os.path.join(os.path.dirname(__file__), "..")
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "0", "--nodes", "2"],
+        capture_output=True, text=True, cwd=repo, timeout=300,
+    )
+    assert out.returncode == 0, out.stderr
+    line = out.stdout.strip().splitlines()[-1]
+    data = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in data
+    assert data["higher_is_better"] is False
+    assert data["scaling"] == "weak"
+    assert data["config"]["model"] == "amdgpu-driver-rolling-upgrade"

@@ -1,0 +1,57 @@
+"""Native GPU validator tests.
+
+The numerics tests compare the HIP MFMA kernels against plain CPU float32
+references (the MFMA f32 path is specified to be a bit-exact fmaf chain).
+All tests here require a real MI355X."""
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def native():
+    from k8s_operator_libs_amd.validation import GpuHealthError, load_native_validator
+
+    mod = load_native_validator()
+    if mod is None:
+        raise GpuHealthError(
+            "native validator must be present on a GPU box - refusing to skip"
+        )
+    return mod
+
+
+def test_device_probe_is_gfx950(native):
+    probe = native.device_probe(0)
+    assert "gfx950" in probe["gcn_arch"]
+    assert probe["warp_size"] == 64
+    assert probe["compute_units"] >= 250  # 256 CUs (some may be harvested)
+    assert probe["hbm_total_gb"] > 250  # 288 GB HBM3E
+
+
+def test_mfma_f32_exact(native):
+    # exact f32 fmaf-chain numerics: error must be ~0
+    assert native.mfma_f32_check(0) <= 1e-6
+
+
+def test_mfma_bf16_close(native):
+    # bf16 inputs, fp32 accumulate; compare against f32 CPU reference over
+    # the same quantized operands
+    assert native.mfma_bf16_check(0) <= 5e-2
+
+
+def test_hbm_bandwidth_sane(native):
+    bw = native.hbm_bandwidth_gbps(0, 512.0, 5)
+    # measured float4 copy on MI355X is ~6.3 TB/s; require at least 2 TB/s
+    assert bw > 2000.0, f"HBM bandwidth suspiciously low: {bw} GB/s"
+
+
+def test_lds_roundtrip(native):
+    assert native.lds_roundtrip_check(0)
+
+
+def test_full_health_check_on_gpu():
+    from k8s_operator_libs_amd.validation import gpu_health_check
+
+    report = gpu_health_check(require_gpu=True)
+    assert report["healthy"], report
