@@ -27,7 +27,7 @@ class SimDaemonSetController:
         self.ds = ds
         self.current_hash = current_hash
         self.ready = ready
-        self._lock = threading.Lock()
+        self._lock = threading.RLock()
         cluster.add_change_hook(self._on_change)
 
     def _on_change(self, event_type, obj):
@@ -72,7 +72,7 @@ class SimMaintenanceOperator:
     def __init__(self, cluster, evict=True):
         self.cluster = cluster
         self.evict = evict
-        self._lock = threading.Lock()
+        self._lock = threading.RLock()
         cluster.add_change_hook(self._on_change)
 
     def _on_change(self, event_type, obj):
