@@ -1,0 +1,252 @@
+"""Mini kube-apiserver: serve a FakeCluster over HTTP.
+
+The wire-level test substrate (our envtest's apiserver): a FastAPI app
+exposing the Kubernetes REST surface the library uses — typed CRUD, JSON
+merge patch, label/field selectors, the pod eviction subresource, and
+API-group discovery — backed by a
+:class:`~k8s_operator_libs_amd.core.fakecluster.FakeCluster`.  The
+:class:`~k8s_operator_libs_amd.core.restclient.RestClient` is tested
+end-to-end against it, and the example operator can run against it for local
+development.
+"""
+
+import json
+import threading
+import time
+from typing import Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse
+
+from .errors import ApiError
+from .fakecluster import FakeCluster
+
+
+def _status_body(exc: ApiError) -> dict:
+    reason = {
+        404: "NotFound",
+        409: "Conflict",
+        400: "BadRequest",
+    }.get(exc.code, "InternalError")
+    if exc.code == 409 and "already exists" in exc.message:
+        reason = "AlreadyExists"
+    return {
+        "kind": "Status",
+        "apiVersion": "v1",
+        "status": "Failure",
+        "message": exc.message,
+        "reason": reason,
+        "code": exc.code,
+    }
+
+
+def create_app(cluster: Optional[FakeCluster] = None):
+    """Build the FastAPI app; returns (app, cluster)."""
+    cluster = cluster or FakeCluster()
+    app = FastAPI(title="amd-k8s-mini-apiserver")
+
+    def _resolve(group: str, version: str, plural: str):
+        api_version = version if not group else f"{group}/{version}"
+        kind = cluster.lookup_by_plural(api_version, plural)
+        if kind is None:
+            raise ApiError(f"resource {plural} not served in {api_version}")
+        return api_version, kind
+
+    def _handle(fn):
+        try:
+            return fn()
+        except ApiError as exc:
+            return JSONResponse(_status_body(exc), status_code=exc.code)
+
+    # -- discovery -----------------------------------------------------------
+
+    @app.get("/api/v1")
+    def discovery_core():
+        return _discovery("", "v1")
+
+    @app.get("/apis/{group}/{version}")
+    def discovery_group(group: str, version: str):
+        return _discovery(group, version)
+
+    def _discovery(group: str, version: str):
+        api_version = version if not group else f"{group}/{version}"
+        resources = []
+        for (av, kind), (plural, namespaced) in cluster._kinds.items():
+            if av == api_version:
+                resources.append(
+                    {"name": plural, "kind": kind, "namespaced": namespaced}
+                )
+        return {"kind": "APIResourceList", "groupVersion": api_version,
+                "resources": resources}
+
+    # -- collection routes ---------------------------------------------------
+
+    async def _list(group, version, plural, request: Request, namespace: str = ""):
+        def run():
+            api_version, kind = _resolve(group, version, plural)
+            items = cluster.list(
+                api_version, kind,
+                namespace=namespace or None,
+                label_selector=request.query_params.get("labelSelector", ""),
+                field_selector=request.query_params.get("fieldSelector", ""),
+            )
+            return {"kind": f"{kind}List", "apiVersion": api_version, "items": items}
+        return _handle(run)
+
+    async def _create(group, version, plural, request: Request, namespace: str = ""):
+        body = json.loads(await request.body())
+
+        def run():
+            _resolve(group, version, plural)
+            if namespace:
+                body.setdefault("metadata", {})["namespace"] = namespace
+            created = cluster.create(body)
+            return JSONResponse(created, status_code=201)
+        return _handle(run)
+
+    async def _get(group, version, plural, name, namespace: str = ""):
+        def run():
+            api_version, kind = _resolve(group, version, plural)
+            return cluster.get(api_version, kind, name, namespace)
+        return _handle(run)
+
+    async def _put(group, version, plural, name, request: Request, namespace: str = ""):
+        body = json.loads(await request.body())
+
+        def run():
+            api_version, kind = _resolve(group, version, plural)
+            if namespace:
+                body.setdefault("metadata", {})["namespace"] = namespace
+            return cluster.update(body)
+        return _handle(run)
+
+    async def _patch(group, version, plural, name, request: Request, namespace: str = ""):
+        body = json.loads(await request.body())
+
+        def run():
+            api_version, kind = _resolve(group, version, plural)
+            return cluster.patch(api_version, kind, name, body, namespace)
+        return _handle(run)
+
+    async def _delete(group, version, plural, name, namespace: str = ""):
+        def run():
+            api_version, kind = _resolve(group, version, plural)
+            cluster.delete(api_version, kind, name, namespace)
+            return {"kind": "Status", "status": "Success"}
+        return _handle(run)
+
+    # core group, cluster-scoped + namespaced
+    app.add_api_route("/api/{version}/{plural}", _wrap_nogroup(_list), methods=["GET"])
+    app.add_api_route("/api/{version}/{plural}", _wrap_nogroup(_create), methods=["POST"])
+    app.add_api_route("/api/{version}/{plural}/{name}", _wrap_nogroup(_get), methods=["GET"])
+    app.add_api_route("/api/{version}/{plural}/{name}", _wrap_nogroup(_put), methods=["PUT"])
+    app.add_api_route("/api/{version}/{plural}/{name}", _wrap_nogroup(_patch), methods=["PATCH"])
+    app.add_api_route("/api/{version}/{plural}/{name}", _wrap_nogroup(_delete), methods=["DELETE"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}",
+                      _wrap_nogroup(_list), methods=["GET"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}",
+                      _wrap_nogroup(_create), methods=["POST"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _wrap_nogroup(_get), methods=["GET"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _wrap_nogroup(_put), methods=["PUT"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _wrap_nogroup(_patch), methods=["PATCH"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _wrap_nogroup(_delete), methods=["DELETE"])
+
+    # named groups
+    app.add_api_route("/apis/{group}/{version}/{plural}", _list, methods=["GET"])
+    app.add_api_route("/apis/{group}/{version}/{plural}", _create, methods=["POST"])
+    app.add_api_route("/apis/{group}/{version}/{plural}/{name}", _get, methods=["GET"])
+    app.add_api_route("/apis/{group}/{version}/{plural}/{name}", _put, methods=["PUT"])
+    app.add_api_route("/apis/{group}/{version}/{plural}/{name}", _patch, methods=["PATCH"])
+    app.add_api_route("/apis/{group}/{version}/{plural}/{name}", _delete, methods=["DELETE"])
+    app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}",
+                      _list, methods=["GET"])
+    app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}",
+                      _create, methods=["POST"])
+    app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _get, methods=["GET"])
+    app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _put, methods=["PUT"])
+    app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _patch, methods=["PATCH"])
+    app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}/{name}",
+                      _delete, methods=["DELETE"])
+
+    # pod eviction subresource
+    @app.post("/api/v1/namespaces/{namespace}/pods/{name}/eviction")
+    async def evict(namespace: str, name: str):
+        def run():
+            cluster.evict_pod(name, namespace)
+            return {"kind": "Status", "status": "Success"}
+        return _handle(run)
+
+    return app, cluster
+
+
+def _wrap_nogroup(fn):
+    """Adapt the group-style handlers to core-group routes (group='')."""
+    import inspect
+
+    params = list(inspect.signature(fn).parameters)
+
+    if "request" in params and "name" in params:
+        async def handler(version: str, plural: str, name: str, request: Request,
+                          namespace: str = ""):
+            return await fn("", version, plural, name, request, namespace)
+    elif "name" in params:
+        async def handler(version: str, plural: str, name: str, namespace: str = ""):
+            return await fn("", version, plural, name, namespace)
+    elif "request" in params:
+        async def handler(version: str, plural: str, request: Request,
+                          namespace: str = ""):
+            return await fn("", version, plural, request, namespace)
+    else:
+        async def handler(version: str, plural: str, namespace: str = ""):
+            return await fn("", version, plural, namespace)
+    return handler
+
+
+class ApiServerHandle:
+    """A running mini-apiserver on a background thread."""
+
+    def __init__(self, server, thread, cluster, url):
+        self._server = server
+        self._thread = thread
+        self.cluster = cluster
+        self.url = url
+
+    def stop(self) -> None:
+        self._server.should_exit = True
+        self._thread.join(timeout=10)
+
+
+def start_apiserver(
+    host: str = "127.0.0.1", port: int = 0,
+    cluster: Optional[FakeCluster] = None,
+) -> ApiServerHandle:
+    """Start the mini-apiserver on a background thread; returns a handle with
+    the bound URL (port=0 picks a free port)."""
+    import socket
+
+    import uvicorn
+
+    if port == 0:
+        with socket.socket() as s:
+            s.bind((host, 0))
+            port = s.getsockname()[1]
+
+    app, cluster = create_app(cluster)
+    config = uvicorn.Config(app, host=host, port=port, log_level="error")
+    server = uvicorn.Server(config)
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+    url = f"http://{host}:{port}"
+    deadline = time.monotonic() + 15
+    while not server.started:
+        if time.monotonic() > deadline:
+            raise RuntimeError("mini-apiserver failed to start in 15s")
+        time.sleep(0.01)
+    return ApiServerHandle(server, thread, cluster, url)
