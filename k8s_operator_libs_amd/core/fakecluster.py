@@ -239,9 +239,11 @@ class FakeCluster:
             obj["metadata"]["resourceVersion"] = self._next_rv()
             obj["metadata"]["generation"] = stored["metadata"].get("generation", 1) + 1
             bucket[(ns, name_)] = obj
+            if self._finalize_if_ready(api_version, kind, ns, name_):
+                # deletion completed by this update: only DELETED was emitted
+                return meta.deep_copy(obj)
             self._notify("MODIFIED", obj)
-            self._finalize_if_ready(api_version, kind, ns, name_)
-            return meta.deep_copy(bucket.get((ns, name_), obj))
+            return meta.deep_copy(obj)
 
     def patch(
         self,
@@ -274,10 +276,10 @@ class FakeCluster:
             meta.json_merge_patch(stored, patch)
             stored["metadata"]["name"] = name  # patches cannot rename
             stored["metadata"]["resourceVersion"] = self._next_rv()
+            if self._finalize_if_ready(api_version, kind, ns, name):
+                return meta.deep_copy(stored)
             self._notify("MODIFIED", stored)
-            self._finalize_if_ready(api_version, kind, ns, name)
-            out = bucket.get((ns, name))
-            return meta.deep_copy(out) if out is not None else {"deleted": True}
+            return meta.deep_copy(stored)
 
     def delete(self, api_version: str, kind: str, name: str, namespace: str = "") -> None:
         with self._lock:
@@ -297,16 +299,19 @@ class FakeCluster:
             del bucket[(ns, name)]
             self._notify("DELETED", stored)
 
-    def _finalize_if_ready(self, api_version: str, kind: str, ns: str, name: str) -> None:
-        """Remove an object whose deletion was pending once finalizers empty."""
+    def _finalize_if_ready(self, api_version: str, kind: str, ns: str, name: str) -> bool:
+        """Remove an object whose deletion was pending once finalizers empty.
+        Returns True if the object was finalized (DELETED emitted)."""
         bucket = self._bucket(api_version, kind)
         stored = bucket.get((ns, name))
         if stored is None:
-            return
+            return False
         md = stored.get("metadata", {})
         if "deletionTimestamp" in md and not (md.get("finalizers") or []):
             del bucket[(ns, name)]
             self._notify("DELETED", stored)
+            return True
+        return False
 
     # -- pods ----------------------------------------------------------------
 

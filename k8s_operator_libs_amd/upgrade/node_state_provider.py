@@ -48,6 +48,8 @@ class NodeUpgradeStateProvider:
         self._client = client
         self._recorder = event_recorder
         self._mutex = util.KeyedMutex()
+        # set by the state manager; transition counts land here
+        self.metrics = None
 
     def get_node(self, name: str) -> K8sObject:
         return self._client.get_node(name)
@@ -78,6 +80,10 @@ class NodeUpgradeStateProvider:
                 )
                 raise
             meta.labels(node)[key] = new_state
+            if self.metrics is not None:
+                self.metrics.state_transitions.inc(old_state, new_state)
+                if new_state == consts.UPGRADE_STATE_FAILED:
+                    self.metrics.upgrade_failures.inc()
             logger.info("node %s upgrade state: %r -> %r", node_name, old_state, new_state)
             log_eventf(
                 self._recorder, node, EVENT_TYPE_NORMAL, util.get_event_reason(),

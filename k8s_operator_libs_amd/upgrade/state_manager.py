@@ -12,10 +12,12 @@ operator's reconcile is simply ``apply_state(build_state(...), policy)``.
 from __future__ import annotations
 
 import logging
+import time
 from dataclasses import dataclass, field
 from typing import Dict, Optional
 
 from ..api.upgrade.v1alpha1 import DriverUpgradePolicySpec
+from ..metrics import MetricsRegistry, default_registry
 from ..core import meta
 from ..core.client import Client
 from ..core.meta import K8sObject
@@ -53,9 +55,12 @@ class ClusterUpgradeStateManager:
         client: Client,
         event_recorder: Optional[object] = None,
         options: Optional[StateOptions] = None,
+        metrics: Optional[MetricsRegistry] = None,
     ) -> None:
         self.opts = options or StateOptions()
+        self.metrics = metrics or default_registry()
         self.common = CommonUpgradeManager(client, event_recorder)
+        self.common.node_state_provider.metrics = self.metrics
         self.inplace = InplaceNodeStateManager(self.common)
         self.requestor: Optional[RequestorNodeStateManager] = None
         if self.opts.requestor.use_maintenance_operator:
@@ -91,6 +96,14 @@ class ClusterUpgradeStateManager:
         """One GET-free snapshot: DaemonSets and driver pods are listed, then
         each pod's node is fetched.  Fails if any driver DaemonSet has
         unscheduled pods (upgrade_state.go:128-131)."""
+        t0 = time.perf_counter()
+        state = self._build_state(namespace, driver_labels)
+        self.metrics.build_state_duration.observe(time.perf_counter() - t0)
+        for state_name, node_states in state.node_states.items():
+            self.metrics.node_states.set(len(node_states), state_name)
+        return state
+
+    def _build_state(self, namespace: str, driver_labels: Dict[str, str]) -> ClusterUpgradeState:
         state = ClusterUpgradeState()
         daemonsets = self.common.get_driver_daemonsets(namespace, driver_labels)
         selector = ",".join(f"{k}={v}" for k, v in sorted(driver_labels.items()))
@@ -135,6 +148,19 @@ class ClusterUpgradeStateManager:
     # -- one state-machine tick (upgrade_state.go:171-281) --------------------
 
     def apply_state(
+        self,
+        current_state: Optional[ClusterUpgradeState],
+        upgrade_policy: Optional[DriverUpgradePolicySpec],
+    ) -> None:
+        t0 = time.perf_counter()
+        try:
+            self._apply_state(current_state, upgrade_policy)
+        finally:
+            dt = time.perf_counter() - t0
+            self.metrics.apply_state_duration.observe(dt)
+            self.metrics.reconcile_duration.observe(dt)
+
+    def _apply_state(
         self,
         current_state: Optional[ClusterUpgradeState],
         upgrade_policy: Optional[DriverUpgradePolicySpec],

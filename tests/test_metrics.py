@@ -1,0 +1,87 @@
+"""Metrics tests: reconcile histograms, transition counters, exposition."""
+
+import pytest
+
+from k8s_operator_libs_amd.metrics import (
+    Counter,
+    Gauge,
+    Histogram,
+    MetricsRegistry,
+)
+from k8s_operator_libs_amd.upgrade import consts
+from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
+
+from builders import DRIVER_LABELS, DRIVER_NS
+from test_state_manager import policy, setup_cluster
+
+
+def test_histogram_quantiles():
+    h = Histogram("h")
+    for v in [0.001 * i for i in range(1, 101)]:
+        h.observe(v)
+    assert h.count == 100
+    assert abs(h.quantile(0.5) - 0.05) < 0.005
+    assert h.quantile(0.99) >= 0.095
+    snap = h.snapshot()
+    assert snap["count"] == 100 and snap["sum"] > 0
+
+
+def test_counter_and_gauge_labels():
+    c = Counter("c")
+    c.inc("a", "b")
+    c.inc("a", "b")
+    c.inc("x", "y")
+    assert c.value("a", "b") == 2
+    g = Gauge("g")
+    g.set(3, "done")
+    assert g.value("done") == 3
+
+
+def test_state_manager_records_metrics(client):
+    reg = MetricsRegistry()
+    manager = ClusterUpgradeStateManager(client, metrics=reg)
+    setup_cluster(client, pod_hash="old", ds_hash="new")
+    state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+    manager.apply_state(state, policy())
+    assert reg.build_state_duration.count == 1
+    assert reg.reconcile_duration.count == 1
+    assert reg.reconcile_duration.quantile(0.5) > 0
+    # "" -> upgrade-required transition counted
+    assert reg.state_transitions.value("", consts.UPGRADE_STATE_UPGRADE_REQUIRED) == 1
+    # gauge recorded the snapshot grouping
+    assert reg.node_states.value("") == 1
+
+
+def test_failure_counter(client):
+    reg = MetricsRegistry()
+    manager = ClusterUpgradeStateManager(client, metrics=reg)
+    setup_cluster(client, node_states=consts.UPGRADE_STATE_POD_RESTART_REQUIRED,
+                  pod_ready=False)
+    pod = client.list_pods(namespace=DRIVER_NS)[0]
+    client.patch("v1", "Pod", pod["metadata"]["name"],
+                 {"status": {"containerStatuses": [
+                     {"name": "driver", "ready": False, "restartCount": 11}]}},
+                 DRIVER_NS)
+    state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+    manager.apply_state(state, policy())
+    assert reg.upgrade_failures.value() == 1
+
+
+def test_prometheus_text_exposition(client):
+    reg = MetricsRegistry()
+    manager = ClusterUpgradeStateManager(client, metrics=reg)
+    setup_cluster(client, pod_hash="old", ds_hash="new")
+    state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+    manager.apply_state(state, policy())
+    text = reg.render_text()
+    assert "amd_upgrade_reconcile_duration_seconds_bucket" in text
+    assert 'le="+Inf"' in text
+    assert "amd_upgrade_state_transitions_total" in text
+    # parseable by the official client if present
+    try:
+        from prometheus_client.parser import text_string_to_metric_families
+    except ImportError:
+        pytest.skip("prometheus_client not installed")
+    families = list(text_string_to_metric_families(text))
+    names = {f.name for f in families}
+    assert "amd_upgrade_reconcile_duration_seconds" in names
