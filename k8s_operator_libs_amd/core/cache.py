@@ -1,0 +1,184 @@
+"""Informer-style cached client.
+
+controller-runtime clients read from a watch-backed cache; writes go to the
+apiserver and the cache catches up asynchronously.  That staleness window is
+why the reference's NodeUpgradeStateProvider polls after every label patch
+(node_upgrade_state_provider.go:92-117).  This module provides the same
+architecture natively:
+
+- :class:`CachedClient` wraps any :class:`~k8s_operator_libs_amd.core.client.Client`
+  that supports ``watch``: reads (get/list) are served from in-memory stores
+  kept current by one watch thread per kind; writes pass through to the
+  delegate.
+- An optional ``sync_delay`` injects artificial cache lag for tests, proving
+  the provider's patch-then-confirm barrier converges rather than
+  double-firing transitions.
+
+``wait_for_cache_sync`` blocks until the initial LIST of every informer has
+been applied (controller-runtime ``WaitForCacheSync`` analogue).
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Dict, List, Optional, Tuple
+
+from . import meta
+from .client import Client
+from .errors import NotFoundError
+from .meta import FieldSelector, K8sObject, LabelSelector
+
+
+class _Informer:
+    def __init__(self, delegate: Client, api_version: str, kind: str,
+                 sync_delay: float = 0.0) -> None:
+        self.api_version = api_version
+        self.kind = kind
+        self._delegate = delegate
+        self._sync_delay = sync_delay
+        self._store: Dict[Tuple[str, str], K8sObject] = {}
+        self._lock = threading.RLock()
+        self._synced = threading.Event()
+        self._stop = threading.Event()
+        self._watch = None
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> None:
+        self._watch = self._delegate.watch(self.api_version, self.kind)
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+
+    def _run(self) -> None:
+        # initial LIST (after the watch opened, so no events are lost)
+        objs = self._delegate.list(self.api_version, self.kind)
+        with self._lock:
+            for obj in objs:
+                self._store[(meta.namespace(obj), meta.name(obj))] = obj
+        self._synced.set()
+        while not self._stop.is_set():
+            item = self._watch.next(timeout=0.2)
+            if item is None:
+                continue
+            if self._sync_delay:
+                time.sleep(self._sync_delay)
+            event_type, obj = item
+            key = (meta.namespace(obj), meta.name(obj))
+            with self._lock:
+                if event_type == "DELETED":
+                    self._store.pop(key, None)
+                else:
+                    current = self._store.get(key)
+                    # resourceVersions are monotonic ints in this stack;
+                    # never regress the cache on out-of-order delivery
+                    if current is not None:
+                        try:
+                            if int(meta.resource_version(obj)) < int(
+                                meta.resource_version(current)
+                            ):
+                                continue
+                        except ValueError:
+                            pass
+                    self._store[key] = obj
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._watch is not None:
+            self._watch.stop()
+        if self._thread is not None:
+            self._thread.join(timeout=5)
+
+    def wait_sync(self, timeout: float = 10.0) -> bool:
+        return self._synced.wait(timeout)
+
+    def get(self, name: str, namespace: str) -> K8sObject:
+        with self._lock:
+            obj = self._store.get((namespace, name))
+            if obj is None:
+                raise NotFoundError(f"{self.kind} {namespace}/{name} not found (cache)")
+            return meta.deep_copy(obj)
+
+    def list(self, namespace: Optional[str], label_selector: str,
+             field_selector: str) -> List[K8sObject]:
+        lsel = LabelSelector(label_selector)
+        fsel = FieldSelector(field_selector)
+        out = []
+        with self._lock:
+            for (ns, _), obj in self._store.items():
+                if namespace not in (None, "") and ns != namespace:
+                    continue
+                if label_selector and not lsel.matches_object(obj):
+                    continue
+                if field_selector and not fsel.matches_object(obj):
+                    continue
+                out.append(meta.deep_copy(obj))
+        out.sort(key=lambda o: (meta.namespace(o), meta.name(o)))
+        return out
+
+
+class CachedClient(Client):
+    """Read-from-cache, write-through client."""
+
+    def __init__(self, delegate: Client, sync_delay: float = 0.0) -> None:
+        self._delegate = delegate
+        self._sync_delay = sync_delay
+        self._informers: Dict[Tuple[str, str], _Informer] = {}
+        self._lock = threading.Lock()
+
+    # expose the underlying cluster when the delegate has one (tests)
+    @property
+    def cluster(self):
+        return getattr(self._delegate, "cluster", None)
+
+    def _informer_for(self, api_version: str, kind: str) -> _Informer:
+        key = (api_version, kind)
+        with self._lock:
+            inf = self._informers.get(key)
+            if inf is None:
+                inf = _Informer(self._delegate, api_version, kind, self._sync_delay)
+                self._informers[key] = inf
+                inf.start()
+        inf.wait_sync()
+        return inf
+
+    def wait_for_cache_sync(self, timeout: float = 10.0) -> bool:
+        with self._lock:
+            informers = list(self._informers.values())
+        return all(inf.wait_sync(timeout) for inf in informers)
+
+    def stop(self) -> None:
+        with self._lock:
+            informers = list(self._informers.values())
+            self._informers.clear()
+        for inf in informers:
+            inf.stop()
+
+    # -- reads: cache --------------------------------------------------------
+
+    def get(self, api_version, kind, name, namespace=""):
+        return self._informer_for(api_version, kind).get(name, namespace)
+
+    def list(self, api_version, kind, namespace=None, label_selector="", field_selector=""):
+        return self._informer_for(api_version, kind).list(
+            namespace, label_selector, field_selector
+        )
+
+    # -- writes: pass-through ------------------------------------------------
+
+    def create(self, obj):
+        return self._delegate.create(obj)
+
+    def update(self, obj):
+        return self._delegate.update(obj)
+
+    def patch(self, api_version, kind, name, patch, namespace=""):
+        return self._delegate.patch(api_version, kind, name, patch, namespace)
+
+    def delete(self, api_version, kind, name, namespace=""):
+        self._delegate.delete(api_version, kind, name, namespace)
+
+    def evict_pod(self, name, namespace):
+        self._delegate.evict_pod(name, namespace)
+
+    def watch(self, api_version, kind):
+        return self._delegate.watch(api_version, kind)

@@ -28,3 +28,12 @@ def _reset_driver_name():
     util.set_driver_name("amdgpu")
     yield
     util.set_driver_name("amdgpu")
+
+
+@pytest.fixture(autouse=True)
+def _reset_metrics():
+    from k8s_operator_libs_amd import metrics
+
+    metrics.reset_default_registry()
+    yield
+    metrics.reset_default_registry()
