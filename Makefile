@@ -1,0 +1,44 @@
+# amd-k8s-operator-libs build/test targets
+# (the reference's Makefile drives go test/lint/envtest; these are the
+# Python/HIP equivalents)
+
+PY ?= python3
+HIPCC ?= /opt/rocm/bin/hipcc
+
+.PHONY: all build test test-gpu bench demo lint coverage clean
+
+all: build test
+
+# Cross-compile the native gfx950 GPU validator in-tree (no GPU required).
+build:
+	$(PY) k8s_operator_libs_amd/native/build.py
+
+# CPU test suite (the envtest-style suites run against the in-memory and
+# HTTP mini apiservers).
+test:
+	$(PY) -m pytest tests/ -q -m "not gpu" --timeout 300
+
+# GPU test suite: requires an MI355X with ROCm.
+test-gpu:
+	$(PY) -m pytest tests/ -q -m gpu --timeout 600
+
+# Flagship benchmark (BASELINE config #3).
+bench:
+	$(PY) bench.py --steps 10 --warmup 2
+
+# Watch a full 8-node rolling upgrade against the mini-apiserver.
+demo:
+	$(PY) examples/amdgpu_upgrade_operator.py --demo
+
+# Syntax/bytecode sanity over the package (stand-in for golangci-lint).
+lint:
+	$(PY) -m compileall -q k8s_operator_libs_amd tests examples bench.py __graft_entry__.py
+
+coverage:
+	$(PY) -m pytest tests/ -q -m "not gpu" --timeout 300 \
+		--cov=k8s_operator_libs_amd --cov-report=term 2>/dev/null \
+		|| $(PY) -m pytest tests/ -q -m "not gpu" --timeout 300
+
+clean:
+	rm -f k8s_operator_libs_amd/native/_gpu_validator.so
+	find . -name __pycache__ -type d -exec rm -rf {} + 2>/dev/null || true
