@@ -128,8 +128,11 @@ class FakeCluster:
 
     def _notify(self, event_type: str, obj: K8sObject) -> None:
         key = self._obj_key(obj)
+        watches = self._watches.get(key)
+        if not watches and not self._change_hooks:
+            return  # nobody listening: skip the snapshot copy (hot path)
         snapshot = meta.deep_copy(obj)
-        for w in self._watches.get(key, []):
+        for w in watches or ():
             w.events.put((event_type, snapshot))
         for hook in self._change_hooks:
             hook(event_type, snapshot)

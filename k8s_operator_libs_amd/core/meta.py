@@ -70,7 +70,32 @@ def controller_owner(obj: K8sObject) -> Optional[Dict[str, Any]]:
 
 
 def deep_copy(obj: K8sObject) -> K8sObject:
-    return copy.deepcopy(obj)
+    """Deep copy of a JSON-shaped tree (dict/list/scalars).
+
+    Hand-rolled: Kubernetes objects are acyclic JSON trees of immutable
+    scalars, so the generic ``copy.deepcopy`` machinery (memo dict, reduce
+    protocol) is pure overhead — this is ~4x faster and the hottest function
+    in the reconcile path (see profiles/).  A native C++ version is used when
+    the ``_jsonops`` extension is available.
+    """
+    return _deep_copy(obj)
+
+
+def _py_deep_copy(obj):
+    t = type(obj)
+    if t is dict:
+        return {k: _py_deep_copy(v) for k, v in obj.items()}
+    if t is list:
+        return [_py_deep_copy(v) for v in obj]
+    return obj
+
+
+try:
+    from ..native._jsonops import deep_copy as _native_deep_copy
+
+    _deep_copy = _native_deep_copy
+except ImportError:  # extension not built yet: pure-Python fallback
+    _deep_copy = _py_deep_copy
 
 
 def dotted_get(obj: K8sObject, path: str, default: Any = None) -> Any:
@@ -217,7 +242,7 @@ def json_merge_patch(target: Any, patch: Any) -> Any:
     deletes a key, everything else replaces.  Returns the patched value
     (mutates dict targets in place)."""
     if not isinstance(patch, dict):
-        return copy.deepcopy(patch)
+        return deep_copy(patch)
     if not isinstance(target, dict):
         target = {}
     for key, value in patch.items():
@@ -226,5 +251,5 @@ def json_merge_patch(target: Any, patch: Any) -> Any:
         elif isinstance(value, dict):
             target[key] = json_merge_patch(target.get(key), value)
         else:
-            target[key] = copy.deepcopy(value)
+            target[key] = deep_copy(value)
     return target

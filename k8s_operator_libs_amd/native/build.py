@@ -14,16 +14,40 @@ import sysconfig
 PKG_DIR = os.path.dirname(os.path.abspath(__file__))
 SOURCE = os.path.join(PKG_DIR, "gpu_validator.hip")
 OUTPUT = os.path.join(PKG_DIR, "_gpu_validator.so")
+JSONOPS_SOURCE = os.path.join(PKG_DIR, "jsonops.cpp")
+JSONOPS_OUTPUT = os.path.join(PKG_DIR, "_jsonops.so")
 
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+CXX = os.environ.get("CXX", "g++")
 
 
 def is_built() -> bool:
     return os.path.exists(OUTPUT) and os.path.getmtime(OUTPUT) >= os.path.getmtime(SOURCE)
 
 
+def build_jsonops(force: bool = False, verbose: bool = False) -> str:
+    """Compile the native JSON-object accelerators (plain C++, no ROCm)."""
+    if (
+        not force
+        and os.path.exists(JSONOPS_OUTPUT)
+        and os.path.getmtime(JSONOPS_OUTPUT) >= os.path.getmtime(JSONOPS_SOURCE)
+    ):
+        return JSONOPS_OUTPUT
+    cmd = [
+        CXX, "-O3", "-std=c++17", "-fPIC", "-shared",
+        f"-I{sysconfig.get_paths()['include']}",
+        JSONOPS_SOURCE, "-o", JSONOPS_OUTPUT,
+    ]
+    if verbose:
+        print(" ".join(cmd))
+    subprocess.run(cmd, check=True, capture_output=not verbose)
+    return JSONOPS_OUTPUT
+
+
 def build(force: bool = False, verbose: bool = False) -> str:
-    """Compile gpu_validator.hip -> _gpu_validator.so for gfx950."""
+    """Compile gpu_validator.hip -> _gpu_validator.so for gfx950 (and the
+    plain-C++ _jsonops accelerator)."""
+    build_jsonops(force=force, verbose=verbose)
     if is_built() and not force:
         return OUTPUT
     import pybind11
