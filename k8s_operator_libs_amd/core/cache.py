@@ -100,8 +100,8 @@ class _Informer:
 
     def list(self, namespace: Optional[str], label_selector: str,
              field_selector: str) -> List[K8sObject]:
-        lsel = LabelSelector(label_selector)
-        fsel = FieldSelector(field_selector)
+        lsel = meta.parse_label_selector(label_selector)
+        fsel = meta.parse_field_selector(field_selector)
         out = []
         with self._lock:
             for (ns, _), obj in self._store.items():

@@ -202,8 +202,8 @@ class FakeCluster:
         label_selector: str = "",
         field_selector: str = "",
     ) -> List[K8sObject]:
-        lsel = LabelSelector(label_selector)
-        fsel = FieldSelector(field_selector)
+        lsel = meta.parse_label_selector(label_selector)
+        fsel = meta.parse_field_selector(field_selector)
         with self._lock:
             bucket = self._bucket(api_version, kind)
             out = []

@@ -9,6 +9,7 @@ JSON-merge-patch used by the client substrate and the upgrade managers.
 from __future__ import annotations
 
 import copy
+import functools as _functools
 import re
 from typing import Any, Dict, Iterable, List, Optional
 
@@ -196,6 +197,18 @@ class LabelSelector:
 
     def matches_object(self, obj: K8sObject) -> bool:
         return self.matches(obj.get("metadata", {}).get("labels", {}) or {})
+
+
+@_functools.lru_cache(maxsize=512)
+def parse_label_selector(selector: str) -> LabelSelector:
+    """Cached selector parse — selectors repeat every reconcile tick and
+    instances are immutable after construction."""
+    return LabelSelector(selector)
+
+
+@_functools.lru_cache(maxsize=512)
+def parse_field_selector(selector: str) -> "FieldSelector":
+    return FieldSelector(selector)
 
 
 def match_labels_selector(match: Dict[str, str]) -> LabelSelector:
