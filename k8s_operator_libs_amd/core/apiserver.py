@@ -82,6 +82,9 @@ def create_app(cluster: Optional[FakeCluster] = None):
     # -- collection routes ---------------------------------------------------
 
     async def _list(group, version, plural, request: Request, namespace: str = ""):
+        if request.query_params.get("watch") in ("true", "1"):
+            return _watch_stream(group, version, plural)
+
         def run():
             api_version, kind = _resolve(group, version, plural)
             items = cluster.list(
@@ -92,6 +95,35 @@ def create_app(cluster: Optional[FakeCluster] = None):
             )
             return {"kind": f"{kind}List", "apiVersion": api_version, "items": items}
         return _handle(run)
+
+    def _watch_stream(group, version, plural):
+        """Kubernetes-style watch: stream newline-delimited WatchEvent JSON."""
+        from fastapi.responses import StreamingResponse
+
+        try:
+            api_version, kind = _resolve(group, version, plural)
+        except ApiError as exc:
+            return JSONResponse(_status_body(exc), status_code=exc.code)
+        watch = cluster.watch(api_version, kind)
+
+        def gen():
+            try:
+                # immediate bookmark: the client knows the watch is
+                # registered server-side before it issues its initial LIST,
+                # closing the lost-event gap (real apiservers close it with
+                # resourceVersion-anchored watches)
+                yield json.dumps({"type": "BOOKMARK", "object": None}) + "\n"
+                while True:
+                    item = watch.next(timeout=0.5)
+                    if item is None:
+                        yield ""  # keep-alive; also surfaces disconnects
+                        continue
+                    event_type, obj = item
+                    yield json.dumps({"type": event_type, "object": obj}) + "\n"
+            finally:
+                watch.stop()
+
+        return StreamingResponse(gen(), media_type="application/json")
 
     async def _create(group, version, plural, request: Request, namespace: str = ""):
         body = json.loads(await request.body())

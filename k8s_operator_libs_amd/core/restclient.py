@@ -213,6 +213,13 @@ class RestClient(Client):
         resp = self._http.post(path, content=json.dumps(body))
         self._raise_for(resp)
 
+    def watch(self, api_version: str, kind: str):
+        """Open a Kubernetes-style watch stream (``?watch=true``), returning
+        an object with ``next(timeout)`` / ``stop()`` like FakeCluster's
+        Watch — so :class:`~k8s_operator_libs_amd.core.cache.CachedClient`
+        runs unchanged over REST."""
+        return _HttpWatch(self, api_version, kind)
+
     # -- discovery (for crdutil.wait_for_crds) ----------------------------------
 
     def discover_resource(self, api_version: str, plural: str) -> bool:
@@ -230,3 +237,63 @@ class RestClient(Client):
 
     def close(self) -> None:
         self._http.close()
+
+
+class _HttpWatch:
+    """Streaming watch over HTTP: a reader thread feeds a queue of
+    ``(event_type, object)`` tuples parsed from newline-delimited
+    WatchEvent JSON."""
+
+    def __init__(self, client: RestClient, api_version: str, kind: str) -> None:
+        import queue
+        import threading
+
+        self._queue: "queue.Queue" = queue.Queue()
+        self._stop = threading.Event()
+        self._connected = threading.Event()
+        path = client._collection_path(api_version, kind, "")
+        url = client.base_url + path
+
+        def reader():
+            try:
+                with httpx.stream(
+                    "GET", url, params={"watch": "true"},
+                    headers=dict(client._http.headers), timeout=None,
+                ) as resp:
+                    for line in resp.iter_lines():
+                        if self._stop.is_set():
+                            break
+                        line = line.strip()
+                        if not line:
+                            continue
+                        try:
+                            event = json.loads(line)
+                        except ValueError:
+                            continue
+                        if event.get("type") == "BOOKMARK":
+                            self._connected.set()
+                            continue
+                        self._queue.put((event.get("type"), event.get("object")))
+            except Exception:
+                if not self._stop.is_set():
+                    raise
+            finally:
+                self._connected.set()  # never leave a waiter hanging
+
+        self._thread = threading.Thread(target=reader, daemon=True)
+        self._thread.start()
+        # Block until the server has registered the watch: events fired
+        # after this point (e.g. by an informer's post-watch initial LIST)
+        # cannot be lost.
+        self._connected.wait(10.0)
+
+    def next(self, timeout: Optional[float] = None):
+        import queue
+
+        try:
+            return self._queue.get(timeout=timeout)
+        except queue.Empty:
+            return None
+
+    def stop(self) -> None:
+        self._stop.set()

@@ -131,3 +131,43 @@ def test_full_upgrade_lifecycle_over_rest(rest, server):
     assert state_of(rest, "node-0") == consts.UPGRADE_STATE_DONE
     pods = rest.list_pods(namespace=DRIVER_NS)
     assert pods[0]["metadata"]["labels"]["controller-revision-hash"] == "new"
+
+
+def test_http_watch_stream(rest, server):
+    w = rest.watch("v1", "Node")
+    import time
+    time.sleep(0.2)  # let the stream connect
+    server.cluster.create({"apiVersion": "v1", "kind": "Node",
+                           "metadata": {"name": "wn1"}, "spec": {}})
+    server.cluster.patch("v1", "Node", "wn1", {"metadata": {"labels": {"s": "1"}}})
+    events = []
+    for _ in range(2):
+        item = w.next(timeout=5)
+        assert item is not None, "watch event not delivered"
+        events.append(item[0])
+    assert events == ["ADDED", "MODIFIED"]
+    w.stop()
+
+
+def test_cached_client_over_rest(rest, server):
+    """Informer cache fed by the HTTP watch stream: the production
+    architecture end-to-end."""
+    import time
+
+    from k8s_operator_libs_amd.core.cache import CachedClient
+
+    server.cluster.create({"apiVersion": "v1", "kind": "Node",
+                           "metadata": {"name": "cn1"}, "spec": {}})
+    cached = CachedClient(rest)
+    try:
+        assert cached.get_node("cn1")["metadata"]["name"] == "cn1"
+        server.cluster.patch("v1", "Node", "cn1",
+                             {"metadata": {"labels": {"x": "1"}}})
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            if cached.get_node("cn1")["metadata"].get("labels", {}).get("x") == "1":
+                break
+            time.sleep(0.01)
+        assert cached.get_node("cn1")["metadata"]["labels"]["x"] == "1"
+    finally:
+        cached.stop()
