@@ -288,6 +288,68 @@ static bool lds_roundtrip_check(int device) {
   return true;
 }
 
+// ---------------------------------------------------------------------------
+// xGMI peer-to-peer probe: on multi-GPU nodes check that every peer of
+// `device` is reachable and measure the p2p copy bandwidth over the xGMI
+// links (7 point-to-point links x ~153 GB/s per MI355X GPU).  The AMD-native
+// analogue of the reference's OFED/NIC validation surface.
+// ---------------------------------------------------------------------------
+
+static py::list xgmi_p2p_probe(int device, double buf_mib, int iters) {
+  py::list out;
+  int count = 0;
+  HIP_CHECK(hipGetDeviceCount(&count));
+  size_t bytes = (size_t)(buf_mib * 1024.0 * 1024.0);
+  for (int peer = 0; peer < count; ++peer) {
+    if (peer == device) continue;
+    py::dict entry;
+    entry["peer"] = peer;
+    int can = 0;
+    HIP_CHECK(hipDeviceCanAccessPeer(&can, device, peer));
+    entry["accessible"] = (bool)can;
+    if (!can) {
+      out.append(entry);
+      continue;
+    }
+    HIP_CHECK(hipSetDevice(device));
+    hipError_t en = hipDeviceEnablePeerAccess(peer, 0);
+    if (en != hipSuccess && en != hipErrorPeerAccessAlreadyEnabled) {
+      entry["error"] = std::string(hipGetErrorString(en));
+      out.append(entry);
+      continue;
+    }
+    void* src = nullptr;
+    void* dst = nullptr;
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipMalloc(&src, bytes));
+    HIP_CHECK(hipMemset(src, 1, bytes));
+    HIP_CHECK(hipSetDevice(peer));
+    HIP_CHECK(hipMalloc(&dst, bytes));
+    HIP_CHECK(hipSetDevice(device));
+    hipEvent_t t0, t1;
+    HIP_CHECK(hipEventCreate(&t0));
+    HIP_CHECK(hipEventCreate(&t1));
+    HIP_CHECK(hipMemcpyPeer(dst, peer, src, device, bytes));  // warmup
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipEventRecord(t0));
+    for (int i = 0; i < iters; ++i)
+      HIP_CHECK(hipMemcpyPeer(dst, peer, src, device, bytes));
+    HIP_CHECK(hipEventRecord(t1));
+    HIP_CHECK(hipEventSynchronize(t1));
+    float ms = 0.f;
+    HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+    entry["bandwidth_gbps"] = ((double)bytes * iters / 1e9) / ((double)ms / 1e3);
+    HIP_CHECK(hipEventDestroy(t0));
+    HIP_CHECK(hipEventDestroy(t1));
+    HIP_CHECK(hipFree(src));
+    HIP_CHECK(hipSetDevice(peer));
+    HIP_CHECK(hipFree(dst));
+    HIP_CHECK(hipSetDevice(device));
+    out.append(entry);
+  }
+  return out;
+}
+
 PYBIND11_MODULE(_gpu_validator, m) {
   m.doc() = "MI355X (gfx950) native GPU health validator";
   m.def("device_probe", &device_probe, py::arg("device") = 0);
@@ -299,4 +361,7 @@ PYBIND11_MODULE(_gpu_validator, m) {
         py::arg("buf_mib") = 1024.0, py::arg("iters") = 10,
         "Streaming float4 copy bandwidth in GB/s (read+write)");
   m.def("lds_roundtrip_check", &lds_roundtrip_check, py::arg("device") = 0);
+  m.def("xgmi_p2p_probe", &xgmi_p2p_probe, py::arg("device") = 0,
+        py::arg("buf_mib") = 256.0, py::arg("iters") = 5,
+        "Peer accessibility + p2p copy bandwidth (GB/s) to every other GPU");
 }

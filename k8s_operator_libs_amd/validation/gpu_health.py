@@ -113,12 +113,21 @@ def gpu_health_check(
     lds_ok = native.lds_roundtrip_check(device)
     report["checks"]["lds_ok"] = lds_ok
 
+    # multi-GPU nodes: verify every xGMI peer link is up
+    xgmi_ok = True
+    if probe.get("device_count", 1) > 1:
+        links = native.xgmi_p2p_probe(device, 64.0, 3)
+        report["checks"]["xgmi_links"] = [dict(l) for l in links]
+        xgmi_ok = all(l.get("accessible") for l in links)
+    report["checks"]["xgmi_ok"] = xgmi_ok
+
     report["healthy"] = bool(
         arch_ok
         and mfma_f32_err <= MFMA_F32_MAX_ERR
         and mfma_bf16_err <= MFMA_BF16_MAX_ERR
         and bw >= HBM_MIN_GBPS
         and lds_ok
+        and xgmi_ok
     )
     return report
 

@@ -55,3 +55,17 @@ def test_full_health_check_on_gpu():
 
     report = gpu_health_check(require_gpu=True)
     assert report["healthy"], report
+
+
+def test_xgmi_p2p_probe(native):
+    import pytest as _pytest
+
+    probe = native.device_probe(0)
+    if probe["device_count"] < 2:
+        _pytest.skip("single-GPU box: no xGMI peers to probe")
+    links = native.xgmi_p2p_probe(0, 64.0, 3)
+    assert len(links) == probe["device_count"] - 1
+    for link in links:
+        assert link["accessible"], f"xGMI peer {link['peer']} unreachable"
+        # each MI355X xGMI link is ~153 GB/s; require a sane floor
+        assert link["bandwidth_gbps"] > 20.0, dict(link)
