@@ -120,3 +120,45 @@ class DriverUpgradePolicySpec(_SpecBase):
     @classmethod
     def _validate_max_unavailable(cls, v):
         return IntOrString.validate(v)
+
+
+def openapi_v3_schema() -> dict:
+    """Structural OpenAPI v3 schema of DriverUpgradePolicySpec for embedding
+    in a consumer CRD (the kubebuilder controller-gen analogue: the
+    reference's consumers get this from the marker comments on
+    upgrade_spec.go).  Defaults and minimums match the field definitions."""
+
+    def prune(schema: dict) -> dict:
+        # pydantic emits $defs/anyOf forms CRDs don't accept; inline and
+        # simplify to the structural-schema subset
+        defs = schema.pop("$defs", {})
+
+        def walk(node):
+            if isinstance(node, dict):
+                if "$ref" in node:
+                    ref = node.pop("$ref").rsplit("/", 1)[-1]
+                    node.update(walk(dict(defs[ref])))
+                if "anyOf" in node:
+                    # int-or-string fields -> x-kubernetes-int-or-string
+                    options = node.pop("anyOf")
+                    types = {o.get("type") for o in options}
+                    if types >= {"integer", "string"}:
+                        node["x-kubernetes-int-or-string"] = True
+                    elif len(options) == 1:
+                        node.update(walk(options[0]))
+                    else:
+                        # optional nested spec (Type | null): take the object arm
+                        arm = next((o for o in options if o.get("type") != "null"),
+                                   options[0])
+                        node.update(walk(dict(arm)))
+                node.pop("title", None)
+                for v in list(node.values()):
+                    walk(v)
+            elif isinstance(node, list):
+                for v in node:
+                    walk(v)
+            return node
+
+        return walk(schema)
+
+    return prune(DriverUpgradePolicySpec.model_json_schema(by_alias=True))

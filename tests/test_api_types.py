@@ -85,3 +85,21 @@ def test_deep_copy_is_independent():
     q = p.deep_copy()
     q.drain_spec.enable = False
     assert p.drain_spec.enable is True
+
+
+def test_openapi_v3_schema_for_crd_embedding():
+    from k8s_operator_libs_amd.api.upgrade.v1alpha1 import openapi_v3_schema
+
+    schema = openapi_v3_schema()
+    props = schema["properties"]
+    assert props["autoUpgrade"]["default"] is False
+    assert props["maxParallelUpgrades"]["default"] == 1
+    assert props["maxParallelUpgrades"]["minimum"] == 0
+    assert props["maxUnavailable"]["x-kubernetes-int-or-string"] is True
+    assert props["maxUnavailable"]["default"] == "25%"
+    assert props["drainSpec"]["properties"]["timeoutSeconds"]["default"] == 300
+    assert props["podDeletion"]["properties"]["timeoutSecond"]["default"] == 300
+    # structural: no $refs/anyOf remain anywhere
+    import json
+    text = json.dumps(schema)
+    assert "$ref" not in text and "anyOf" not in text
