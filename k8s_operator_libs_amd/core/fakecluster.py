@@ -80,6 +80,10 @@ class FakeCluster:
         self._kinds = dict(_BUILTIN_KINDS)
         self._watches: Dict[Tuple[str, str], List[Watch]] = {}
         self._change_hooks: List[Callable[[str, K8sObject], None]] = []
+        # spec.nodeName index for pods: nodeName -> {(ns, name)}; keeps
+        # per-node pod LISTs (the reconcile loop's hottest query) O(pods on
+        # node) instead of O(all pods)
+        self._pod_node_index: Dict[str, set] = {}
 
     # -- kind registry -------------------------------------------------------
 
@@ -137,6 +141,18 @@ class FakeCluster:
         for hook in self._change_hooks:
             hook(event_type, snapshot)
 
+    def _index_pod(self, obj: K8sObject, remove: bool = False) -> None:
+        if meta.kind(obj) != "Pod":
+            return
+        node = obj.get("spec", {}).get("nodeName", "")
+        key = (meta.namespace(obj), meta.name(obj))
+        if remove:
+            bucket = self._pod_node_index.get(node)
+            if bucket is not None:
+                bucket.discard(key)
+        else:
+            self._pod_node_index.setdefault(node, set()).add(key)
+
     def add_change_hook(self, hook: Callable[[str, K8sObject], None]) -> None:
         """Register a callback fired on every mutation (used by simulated
         kubelets / maintenance operators in tests and benchmarks)."""
@@ -178,6 +194,7 @@ class FakeCluster:
             md.setdefault("creationTimestamp", _now_iso())
             md.setdefault("generation", 1)
             bucket[(ns, name_)] = obj
+            self._index_pod(obj)
             if kind == "CustomResourceDefinition":
                 self.register_crd(obj)
                 self._establish_crd(obj)
@@ -206,8 +223,15 @@ class FakeCluster:
         fsel = meta.parse_field_selector(field_selector)
         with self._lock:
             bucket = self._bucket(api_version, kind)
+            candidates = bucket.items()
+            if kind == "Pod" and field_selector:
+                for fkey, fop, fval in fsel._reqs:
+                    if fkey == "spec.nodeName" and fop == "=":
+                        keys = self._pod_node_index.get(fval, set())
+                        candidates = [(k, bucket[k]) for k in keys if k in bucket]
+                        break
             out = []
-            for (ns, _name), obj in bucket.items():
+            for (ns, _name), obj in candidates:
                 if namespace is not None and namespace != "" and ns != namespace:
                     continue
                 if label_selector and not lsel.matches_object(obj):
@@ -241,7 +265,9 @@ class FakeCluster:
                 obj["metadata"]["deletionTimestamp"] = stored["metadata"]["deletionTimestamp"]
             obj["metadata"]["resourceVersion"] = self._next_rv()
             obj["metadata"]["generation"] = stored["metadata"].get("generation", 1) + 1
+            self._index_pod(stored, remove=True)
             bucket[(ns, name_)] = obj
+            self._index_pod(obj)
             if self._finalize_if_ready(api_version, kind, ns, name_):
                 # deletion completed by this update: only DELETED was emitted
                 return meta.deep_copy(obj)
@@ -276,9 +302,13 @@ class FakeCluster:
                 raise ConflictError(
                     f"{kind} {ns}/{name}: resourceVersion {patch_rv} is stale"
                 )
+            old_node = stored.get("spec", {}).get("nodeName", "") if kind == "Pod" else None
             meta.json_merge_patch(stored, patch)
             stored["metadata"]["name"] = name  # patches cannot rename
             stored["metadata"]["resourceVersion"] = self._next_rv()
+            if kind == "Pod" and stored.get("spec", {}).get("nodeName", "") != old_node:
+                self._pod_node_index.get(old_node, set()).discard((ns, name))
+                self._index_pod(stored)
             if self._finalize_if_ready(api_version, kind, ns, name):
                 return meta.deep_copy(stored)
             self._notify("MODIFIED", stored)
@@ -299,6 +329,7 @@ class FakeCluster:
                     stored["metadata"]["resourceVersion"] = self._next_rv()
                     self._notify("MODIFIED", stored)
                 return
+            self._index_pod(stored, remove=True)
             del bucket[(ns, name)]
             self._notify("DELETED", stored)
 
@@ -311,6 +342,7 @@ class FakeCluster:
             return False
         md = stored.get("metadata", {})
         if "deletionTimestamp" in md and not (md.get("finalizers") or []):
+            self._index_pod(stored, remove=True)
             del bucket[(ns, name)]
             self._notify("DELETED", stored)
             return True
