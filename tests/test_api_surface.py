@@ -1,0 +1,155 @@
+"""API-surface contract: every component in SURVEY.md §2's inventory has a
+working counterpart here.  This test enumerates the mapping explicitly so
+capability parity is machine-checked, not just claimed."""
+
+import inspect
+
+
+def test_api_types_surface():
+    from k8s_operator_libs_amd.api.upgrade import v1alpha1 as api
+
+    # upgrade_spec.go:27-110
+    for cls in ("DriverUpgradePolicySpec", "DrainSpec", "PodDeletionSpec",
+                "WaitForCompletionSpec"):
+        assert hasattr(api, cls)
+    # zz_generated.deepcopy.go
+    assert callable(api.DriverUpgradePolicySpec().deep_copy)
+    # kubebuilder schema markers -> structural schema
+    assert callable(api.openapi_v3_schema)
+
+
+def test_upgrade_constants_surface():
+    from k8s_operator_libs_amd.upgrade import consts
+
+    # consts.go:48-83 — all 13 states
+    assert len(consts.ALL_STATES) == 13
+    # consts.go:20-47 — all 9 key formats, on the amd.com domain
+    fmts = [getattr(consts, n) for n in dir(consts) if n.endswith("_FMT")
+            and "FIELD_SELECTOR" not in n]
+    assert len(fmts) == 9
+    assert all(f.startswith("amd.com/") for f in fmts)
+
+
+def test_util_surface():
+    from k8s_operator_libs_amd.upgrade import util
+
+    # util.go:29-176
+    for fn in ("set_driver_name", "get_driver_name", "get_upgrade_state_label_key",
+               "get_upgrade_skip_node_label_key", "get_upgrade_skip_drain_pod_selector",
+               "get_upgrade_wait_for_safe_driver_load_annotation_key",
+               "get_upgrade_initial_state_annotation_key",
+               "get_wait_for_pod_completion_start_time_annotation_key",
+               "get_validation_start_time_annotation_key",
+               "get_upgrade_requested_annotation_key",
+               "get_upgrade_requestor_mode_annotation_key", "get_event_reason"):
+        assert callable(getattr(util, fn)), fn
+    assert util.StringSet and util.KeyedMutex
+
+
+def test_manager_interfaces_surface():
+    from k8s_operator_libs_amd import upgrade
+
+    # L3 managers + provider
+    assert upgrade.CordonManager and upgrade.DrainManager and upgrade.PodManager
+    assert upgrade.ValidationManager and upgrade.SafeDriverLoadManager
+    assert upgrade.NodeUpgradeStateProvider
+    # L4
+    assert upgrade.ClusterUpgradeStateManager and upgrade.CommonUpgradeManager
+    assert upgrade.InplaceNodeStateManager and upgrade.RequestorNodeStateManager
+    # facade methods (upgrade_state.go:35-53, 329-350)
+    mgr = upgrade.ClusterUpgradeStateManager
+    for method in ("build_state", "apply_state", "with_pod_deletion_enabled",
+                   "with_validation_enabled"):
+        assert callable(getattr(mgr, method)), method
+    # CommonUpgradeStateManager iface (common_manager.go:23-41)
+    common = upgrade.CommonUpgradeManager
+    for method in ("get_total_managed_nodes", "get_upgrades_in_progress",
+                   "get_upgrades_done", "get_upgrades_failed",
+                   "get_upgrades_pending", "get_upgrades_available",
+                   "get_current_unavailable_nodes", "is_pod_deletion_enabled",
+                   "is_validation_enabled", "skip_node_upgrade",
+                   "is_upgrade_requested"):
+        assert callable(getattr(common, method)), method
+    # ProcessNodeStateManager iface (common_manager.go:47-54)
+    for impl in (upgrade.InplaceNodeStateManager, upgrade.RequestorNodeStateManager):
+        for method in ("process_upgrade_required_nodes",
+                       "process_node_maintenance_required_nodes",
+                       "process_uncordon_required_nodes"):
+            assert callable(getattr(impl, method)), (impl, method)
+
+
+def test_requestor_surface():
+    from k8s_operator_libs_amd.upgrade import requestor
+
+    # upgrade_requestor.go:40-109, 527-551
+    assert requestor.RequestorOptions and requestor.get_requestor_opts_from_envs
+    assert requestor.MAINTENANCE_OP_EVICTION_GPU == "amd.com/gpu-*"
+    assert requestor.MAINTENANCE_OP_EVICTION_RDMA == "amd.com/rdma*"
+    assert requestor.DEFAULT_NODE_MAINTENANCE_NAME_PREFIX == "amd-operator"
+    assert callable(requestor.requestor_id_predicate)
+    assert callable(requestor.condition_changed_predicate)
+    assert callable(requestor.RequestorNodeStateManager.set_default_node_maintenance)
+
+
+def test_crdutil_surface():
+    from k8s_operator_libs_amd import crdutil
+
+    # crdutil.go:44-319
+    for fn in ("process_crds", "apply_crds", "delete_crds", "wait_for_crds",
+               "walk_crd_paths", "parse_crds_from_paths"):
+        assert callable(getattr(crdutil, fn)), fn
+    assert crdutil.CRD_OPERATION_APPLY == "apply"
+    assert crdutil.CRD_OPERATION_DELETE == "delete"
+
+
+def test_consts_surface():
+    from k8s_operator_libs_amd import consts
+
+    # pkg/consts/consts.go:24-29
+    assert consts.LOG_LEVEL_ERROR == -2
+    assert consts.LOG_LEVEL_WARNING == -1
+    assert consts.LOG_LEVEL_INFO == 0
+    assert consts.LOG_LEVEL_DEBUG == 1
+
+
+def test_mocks_surface():
+    from k8s_operator_libs_amd.upgrade import mocks
+
+    # pkg/upgrade/mocks (5 mockery files) + safe-load double
+    for cls in ("MockCordonManager", "MockDrainManager",
+                "MockNodeUpgradeStateProvider", "MockPodManager",
+                "MockValidationManager", "MockSafeDriverLoadManager"):
+        assert inspect.isclass(getattr(mocks, cls)), cls
+
+
+def test_core_substrate_surface():
+    from k8s_operator_libs_amd import core
+    from k8s_operator_libs_amd.core import apiserver, cache, restclient
+
+    assert core.Client and core.FakeClient and core.FakeCluster
+    assert core.EventRecorder and core.FakeRecorder
+    assert restclient.RestClient and callable(restclient.RestClient.from_environment)
+    assert cache.CachedClient
+    assert callable(apiserver.start_apiserver)
+
+
+def test_validation_surface():
+    from k8s_operator_libs_amd import validation
+
+    assert callable(validation.gpu_health_check)
+    assert callable(validation.smi_probe)
+    assert callable(validation.load_native_validator)
+
+
+def test_event_recorder_writes_events(client):
+    from k8s_operator_libs_amd.core.events import EventRecorder
+
+    node = {"apiVersion": "v1", "kind": "Node", "metadata": {"name": "n1"}}
+    client.create(dict(node, spec={}))
+    rec = EventRecorder(client)
+    rec.eventf(node, "Normal", "AMDGPUDriverUpgrade", "state changed to '{}'", "upgrade-done")
+    events = client.list("v1", "Event", namespace="default")
+    assert len(events) == 1
+    assert events[0]["reason"] == "AMDGPUDriverUpgrade"
+    assert "upgrade-done" in events[0]["message"]
+    assert events[0]["involvedObject"]["name"] == "n1"
