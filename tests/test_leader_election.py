@@ -1,0 +1,67 @@
+"""Lease-based leader election tests."""
+
+import threading
+import time
+
+from k8s_operator_libs_amd.core.leaderelection import LeaderElector
+
+
+def test_single_candidate_acquires_and_releases(client):
+    client.cluster.register_kind("coordination.k8s.io/v1", "Lease", "leases", True)
+    elector = LeaderElector(client, "amd-gpu-operator", identity="a",
+                            lease_duration=2.0, retry_period=0.05)
+    ran = threading.Event()
+
+    def lead():
+        ran.set()
+        while elector.is_leading() and not elector._stop.is_set():
+            time.sleep(0.02)
+
+    t = threading.Thread(target=lambda: elector.run(lead), daemon=True)
+    t.start()
+    assert ran.wait(5.0)
+    assert elector.is_leading()
+    lease = client.get("coordination.k8s.io/v1", "Lease", "amd-gpu-operator", "default")
+    assert lease["spec"]["holderIdentity"] == "a"
+    elector.stop()
+    t.join(timeout=5)
+    lease = client.get("coordination.k8s.io/v1", "Lease", "amd-gpu-operator", "default")
+    assert not lease["spec"].get("holderIdentity")
+
+
+def test_second_candidate_waits_then_takes_over(client):
+    client.cluster.register_kind("coordination.k8s.io/v1", "Lease", "leases", True)
+    a = LeaderElector(client, "op", identity="a", lease_duration=0.5, retry_period=0.05)
+    b = LeaderElector(client, "op", identity="b", lease_duration=0.5, retry_period=0.05)
+    a_leading = threading.Event()
+    b_leading = threading.Event()
+
+    def lead_a():
+        a_leading.set()
+        while a.is_leading() and not a._stop.is_set():
+            time.sleep(0.02)
+
+    def lead_b():
+        b_leading.set()
+        while b.is_leading() and not b._stop.is_set():
+            time.sleep(0.02)
+
+    ta = threading.Thread(target=lambda: a.run(lead_a), daemon=True)
+    ta.start()
+    assert a_leading.wait(5.0)
+    tb = threading.Thread(target=lambda: b.run(lead_b), daemon=True)
+    tb.start()
+    # b must not lead while a renews
+    time.sleep(0.4)
+    assert not b_leading.is_set()
+    # a dies without releasing (simulated crash: stop renewing, hold lease)
+    a._leading.clear()
+    a._stop.set()
+    # b takes over after the lease expires
+    assert b_leading.wait(5.0)
+    lease = client.get("coordination.k8s.io/v1", "Lease", "op", "default")
+    assert lease["spec"]["holderIdentity"] == "b"
+    assert lease["spec"]["leaseTransitions"] >= 1
+    b.stop()
+    ta.join(timeout=5)
+    tb.join(timeout=5)
