@@ -169,10 +169,17 @@ def _run_gpu_validation(device):
 
 def run_rolling_upgrade_benchmark(
     n_nodes=8, steps=10, warmup=2, max_parallel=2, gpu_validate=False,
-    device=0, gpu_pods_per_node=2, print_json=True,
+    device=0, gpu_pods_per_node=2, print_json=True, converge=True,
 ):
     """Run `warmup` untimed + `steps` timed full rolling upgrades; returns a
-    result dict (single-process path; bench main() adds distribution)."""
+    result dict (single-process path; bench main() adds distribution).
+
+    ``converge=True`` uses the manager's converging reconcile (nodes travel
+    every synchronously-completable transition per call); ``False`` uses the
+    reference's one-transition-per-tick semantics.
+    """
+    from k8s_operator_libs_amd.metrics import MetricsRegistry
+
     policy = DriverUpgradePolicySpec.model_validate({
         "autoUpgrade": True,
         "maxParallelUpgrades": max_parallel,
@@ -192,21 +199,18 @@ def run_rolling_upgrade_benchmark(
         client = FakeClient()
         ds = _make_cluster(client, n_nodes, "oldrev", "newrev", gpu_pods_per_node)
         _DsController(client.cluster, ds, "newrev")
+        registry = MetricsRegistry()
         manager = (
-            ClusterUpgradeStateManager(client)
+            ClusterUpgradeStateManager(client, metrics=registry)
             .with_pod_deletion_enabled(gpu_pod_deletion_filter)
             .with_validation_enabled(VALIDATOR_SELECTOR)
         )
         t0 = time.perf_counter()
-        ticks = 0
+        rounds = 0
         while True:
-            tick_start = time.perf_counter()
-            state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
-            manager.apply_state(state, policy)
-            manager.wait_idle()
-            if timed:
-                tick_times.append(time.perf_counter() - tick_start)
-            ticks += 1
+            state = manager.reconcile(DRIVER_NS, DRIVER_LABELS, policy,
+                                      converge=converge)
+            rounds += 1
             # validation pods: run the real GPU check, then mark Ready
             for ns_ in state.nodes_in(consts.UPGRADE_STATE_VALIDATION_REQUIRED):
                 node_name = ns_.node["metadata"]["name"]
@@ -224,9 +228,12 @@ def run_rolling_upgrade_benchmark(
             )
             if done == n_nodes:
                 break
-            if ticks > 50 * n_nodes:
-                raise RuntimeError(f"upgrade did not converge after {ticks} ticks")
+            if rounds > 60 * n_nodes:
+                raise RuntimeError(f"upgrade did not converge after {rounds} rounds")
         elapsed = time.perf_counter() - t0
+        ticks = registry.reconcile_duration.count
+        if timed:
+            tick_times.extend(registry.reconcile_duration.samples())
         if timed:
             wall_times.append(elapsed)
             ticks_per_upgrade.append(ticks)
@@ -260,6 +267,8 @@ def main():
     parser.add_argument("--nodes", type=int, default=8, help="simulated nodes per cluster")
     parser.add_argument("--max-parallel", type=int, default=2)
     parser.add_argument("--no-gpu-validate", action="store_true")
+    parser.add_argument("--no-converge", action="store_true",
+                        help="reference semantics: one state transition per reconcile tick")
     args = parser.parse_args()
 
     import torch
@@ -286,18 +295,19 @@ def main():
         if have_cuda:
             torch.cuda.synchronize()
 
+    converge = not args.no_converge
     # warmup (untimed) + timed section bracketed by barrier+sync on both sides
     run_rolling_upgrade_benchmark(
         n_nodes=args.nodes, steps=0, warmup=args.warmup,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
-        device=local_rank if have_cuda else 0, print_json=False,
+        device=local_rank if have_cuda else 0, print_json=False, converge=converge,
     )
     barrier_sync()
     t0 = time.perf_counter()
     result = run_rolling_upgrade_benchmark(
         n_nodes=args.nodes, steps=args.steps, warmup=0,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
-        device=local_rank if have_cuda else 0, print_json=False,
+        device=local_rank if have_cuda else 0, print_json=False, converge=converge,
     )
     barrier_sync()
     elapsed = time.perf_counter() - t0
@@ -337,6 +347,7 @@ def main():
                 "podDeletion": True,
                 "gpu_pods_per_node": 2,
                 "gpu_validation": gpu_validate,
+                "converging_reconcile": converge,
                 "global_batch": args.nodes * int(os.environ.get("WORLD_SIZE", "1")),
                 "parallelism": f"dp{world} (one independent simulated "
                                f"{args.nodes}-node cluster per rank)",

@@ -48,6 +48,11 @@ class Histogram:
             if len(self._reservoir) > 4096:
                 self._reservoir = self._reservoir[-2048:]
 
+    def samples(self) -> List[float]:
+        """Recent raw observations (up to the reservoir size)."""
+        with self._lock:
+            return list(self._reservoir)
+
     def quantile(self, q: float) -> float:
         with self._lock:
             if not self._reservoir:

@@ -61,6 +61,20 @@ class ClusterUpgradeState:
     def add(self, state: str, node_state: NodeUpgradeState) -> None:
         self.node_states.setdefault(state, []).append(node_state)
 
+    def regroup(self, state_label_key: str) -> None:
+        """Re-bucket every node by its CURRENT state label.
+
+        Node objects are mutated in place by the state provider, so after a
+        phase runs, regrouping makes nodes visible to later phases of the
+        same pass.  Used by the live (pipelined) apply mode — the reference
+        always processes the fixed snapshot (one transition per tick)."""
+        from ..core import meta as _meta
+
+        all_states = [ns for lst in self.node_states.values() for ns in lst]
+        self.node_states = {}
+        for node_state in all_states:
+            self.add(_meta.get_label(node_state.node, state_label_key), node_state)
+
 
 def is_orphaned_pod(pod: K8sObject) -> bool:
     return not meta.owner_references(pod)

@@ -502,3 +502,42 @@ class TestCounts:
         counts = manager.counts(state)
         assert counts == {"total": 4, "in_progress": 2, "done": 1,
                           "failed": 1, "pending": 1}
+
+
+class TestLiveReconcile:
+    def test_live_mode_pipelines_multiple_transitions_per_pass(self, client):
+        """Beyond-parity: with converge=True (live regrouping) a full
+        single-node upgrade completes in a handful of reconcile calls, with
+        every label transition still firing exactly once."""
+        from k8s_operator_libs_amd.metrics import MetricsRegistry
+
+        reg = MetricsRegistry()
+        manager = ClusterUpgradeStateManager(client, metrics=reg)
+        ds, _ = setup_cluster(client, pod_hash="old", ds_hash="new")
+        SimDaemonSetController(client.cluster, ds, current_hash="new")
+        pol = policy(maxParallelUpgrades=1, maxUnavailable="100%")
+        calls = 0
+        for _ in range(4):
+            manager.reconcile(DRIVER_NS, DRIVER_LABELS, pol, converge=True)
+            calls += 1
+            if state_of(client, "node-0") == consts.UPGRADE_STATE_DONE:
+                break
+        assert state_of(client, "node-0") == consts.UPGRADE_STATE_DONE
+        assert calls <= 4
+        for (frm, to), count in reg.state_transitions.items().items():
+            assert count == 1, f"{frm}->{to} fired {count} times"
+
+    def test_live_and_reference_modes_reach_same_final_state(self, client):
+        ds, _ = setup_cluster(client, n_nodes=3, pod_hash="old", ds_hash="new")
+        SimDaemonSetController(client.cluster, ds, current_hash="new")
+        manager = ClusterUpgradeStateManager(client)
+        pol = policy(maxParallelUpgrades=2, maxUnavailable="100%",
+                     drainSpec={"enable": True})
+        for _ in range(30):
+            manager.reconcile(DRIVER_NS, DRIVER_LABELS, pol, converge=True)
+            if all(state_of(client, f"node-{i}") == consts.UPGRADE_STATE_DONE
+                   for i in range(3)):
+                break
+        for i in range(3):
+            assert state_of(client, f"node-{i}") == consts.UPGRADE_STATE_DONE
+            assert not client.get_node(f"node-{i}")["spec"].get("unschedulable")
