@@ -27,7 +27,7 @@ from typing import Dict, List, Optional, Tuple
 from . import meta
 from .client import Client
 from .errors import NotFoundError
-from .meta import FieldSelector, K8sObject, LabelSelector
+from .meta import K8sObject
 
 
 class _Informer:

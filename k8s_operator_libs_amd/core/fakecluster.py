@@ -19,7 +19,7 @@ import queue
 import threading
 import time
 import uuid
-from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 from . import meta
 from .errors import (
@@ -28,7 +28,7 @@ from .errors import (
     ConflictError,
     NotFoundError,
 )
-from .meta import FieldSelector, K8sObject, LabelSelector
+from .meta import K8sObject
 
 # (apiVersion, kind) -> (plural, namespaced)
 _BUILTIN_KINDS: Dict[Tuple[str, str], Tuple[str, bool]] = {

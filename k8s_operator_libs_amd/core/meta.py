@@ -8,7 +8,6 @@ JSON-merge-patch used by the client substrate and the upgrade managers.
 
 from __future__ import annotations
 
-import copy
 import functools as _functools
 import re
 from typing import Any, Dict, Iterable, List, Optional

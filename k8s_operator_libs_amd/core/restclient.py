@@ -19,7 +19,7 @@ from __future__ import annotations
 
 import json
 import os
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, Optional
 
 import httpx
 
@@ -32,7 +32,6 @@ from .errors import (
     ConflictError,
     NotFoundError,
 )
-from .meta import K8sObject
 
 SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
 

@@ -28,7 +28,7 @@ import time
 from dataclasses import dataclass, field
 from typing import Callable, List, Optional
 
-from ..api.upgrade.v1alpha1 import DrainSpec, PodDeletionSpec
+from ..api.upgrade.v1alpha1 import DrainSpec
 from ..core import meta
 from ..core.client import Client
 from ..core.errors import NotFoundError

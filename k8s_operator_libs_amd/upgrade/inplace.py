@@ -10,8 +10,6 @@ uncordons and completes nodes not owned by requestor mode.
 from __future__ import annotations
 
 import logging
-from typing import Optional
-
 from ..api.upgrade.v1alpha1 import DriverUpgradePolicySpec, IntOrString
 from ..core import meta
 from . import consts, util

@@ -16,8 +16,6 @@ driver loads on a quiesced node.
 from __future__ import annotations
 
 import logging
-from typing import Optional
-
 from ..core import meta
 from ..core.meta import K8sObject
 from . import consts, util

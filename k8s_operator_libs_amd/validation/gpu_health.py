@@ -7,7 +7,7 @@ import json
 import os
 import shutil
 import subprocess
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 # Acceptance thresholds for an MI355X node.
 MFMA_F32_MAX_ERR = 1e-6       # exact fmaf-chain numerics: effectively zero
