@@ -138,11 +138,15 @@ def run_demo(args):
 
     threading.Thread(target=validator_sim, daemon=True).start()
 
+    from k8s_operator_libs_amd.upgrade.controller import UpgradeController
+
     t0 = time.time()
-    ok = reconcile_loop(
-        manager, client, policy, bench_mod.DRIVER_NS, bench_mod.DRIVER_LABELS,
-        interval=args.interval, max_iterations=200,
+    controller = UpgradeController(
+        manager, bench_mod.DRIVER_NS, bench_mod.DRIVER_LABELS, policy,
+        resync_seconds=max(args.interval, 0.05),
     )
+    ok = controller.run(until_all_done=True, max_reconciles=200)
+    controller.stop()
     log.info("demo %s in %.2fs; metrics: curl 127.0.0.1:%d/metrics",
              "completed" if ok else "DID NOT complete", time.time() - t0,
              args.metrics_port)
