@@ -153,6 +153,69 @@ class _DsController:
             })
 
 
+class _MaintenanceOperatorSim:
+    """Minimal maintenance operator for requestor-mode benchmarking: drives
+    NodeMaintenance objects to Ready (cordon + evict) and honours deletion
+    via a finalizer (uncordon on release)."""
+
+    FINALIZER = "maintenance.amd.com/guard"
+
+    def __init__(self, cluster):
+        import threading
+
+        self.cluster = cluster
+        self._lock = threading.RLock()
+        cluster.add_change_hook(self._on_change)
+
+    def _on_change(self, event_type, obj):
+        if obj.get("kind") != "NodeMaintenance" or event_type not in ("ADDED", "MODIFIED"):
+            return
+        with self._lock:
+            api = "maintenance.amd.com/v1alpha1"
+            try:
+                live = self.cluster.get(api, "NodeMaintenance",
+                                        obj["metadata"]["name"],
+                                        obj["metadata"].get("namespace", ""))
+            except Exception:
+                return
+            node = live.get("spec", {}).get("nodeName", "")
+            if "deletionTimestamp" in live["metadata"]:
+                if node:
+                    self.cluster.patch("v1", "Node", node,
+                                       {"spec": {"unschedulable": None}})
+                self.cluster.patch(api, "NodeMaintenance", live["metadata"]["name"],
+                                   {"metadata": {"finalizers": []}},
+                                   live["metadata"].get("namespace", ""))
+                return
+            fins = live["metadata"].get("finalizers") or []
+            if self.FINALIZER not in fins:
+                self.cluster.patch(api, "NodeMaintenance", live["metadata"]["name"],
+                                   {"metadata": {"finalizers": fins + [self.FINALIZER]}},
+                                   live["metadata"].get("namespace", ""))
+            conds = live.get("status", {}).get("conditions") or []
+            if not any(c.get("type") == "Ready" for c in conds):
+                if node:
+                    self.cluster.patch("v1", "Node", node,
+                                       {"spec": {"unschedulable": True}})
+                    for pod in self.cluster.list("v1", "Pod",
+                                                 field_selector=f"spec.nodeName={node}"):
+                        refs = pod["metadata"].get("ownerReferences") or []
+                        if refs and refs[0].get("kind") == "DaemonSet":
+                            continue
+                        if pod["metadata"].get("labels", {}).get("app") == "amd-gpu-validator":
+                            continue
+                        try:
+                            self.cluster.delete("v1", "Pod", pod["metadata"]["name"],
+                                                pod["metadata"].get("namespace", ""))
+                        except Exception:
+                            pass
+                self.cluster.patch(api, "NodeMaintenance", live["metadata"]["name"],
+                                   {"status": {"conditions": [
+                                       {"type": "Ready", "status": "True",
+                                        "reason": "Ready"}]}},
+                                   live["metadata"].get("namespace", ""))
+
+
 def _run_gpu_validation(device):
     """The validation-pod payload: native gfx950 MFMA smoke on this rank's
     GPU.  Raises if the native extension is missing on a GPU box."""
@@ -170,6 +233,7 @@ def _run_gpu_validation(device):
 def run_rolling_upgrade_benchmark(
     n_nodes=8, steps=10, warmup=2, max_parallel=2, gpu_validate=False,
     device=0, gpu_pods_per_node=2, print_json=True, converge=True,
+    mode="inplace",
 ):
     """Run `warmup` untimed + `steps` timed full rolling upgrades; returns a
     result dict (single-process path; bench main() adds distribution).
@@ -179,6 +243,8 @@ def run_rolling_upgrade_benchmark(
     reference's one-transition-per-tick semantics.
     """
     from k8s_operator_libs_amd.metrics import MetricsRegistry
+    from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
+    from k8s_operator_libs_amd.upgrade.state_manager import StateOptions
 
     policy = DriverUpgradePolicySpec.model_validate({
         "autoUpgrade": True,
@@ -200,8 +266,16 @@ def run_rolling_upgrade_benchmark(
         ds = _make_cluster(client, n_nodes, "oldrev", "newrev", gpu_pods_per_node)
         _DsController(client.cluster, ds, "newrev")
         registry = MetricsRegistry()
+        options = None
+        if mode == "requestor":
+            _MaintenanceOperatorSim(client.cluster)
+            options = StateOptions(requestor=RequestorOptions(
+                use_maintenance_operator=True,
+                requestor_id="amd.gpu.operator",
+                namespace="default",
+            ))
         manager = (
-            ClusterUpgradeStateManager(client, metrics=registry)
+            ClusterUpgradeStateManager(client, metrics=registry, options=options)
             .with_pod_deletion_enabled(gpu_pod_deletion_filter)
             .with_validation_enabled(VALIDATOR_SELECTOR)
         )
@@ -269,6 +343,8 @@ def main():
     parser.add_argument("--no-gpu-validate", action="store_true")
     parser.add_argument("--no-converge", action="store_true",
                         help="reference semantics: one state transition per reconcile tick")
+    parser.add_argument("--mode", choices=["inplace", "requestor"], default="inplace",
+                        help="in-place node ops vs maintenance-operator delegation")
     args = parser.parse_args()
 
     import torch
@@ -301,6 +377,7 @@ def main():
         n_nodes=args.nodes, steps=0, warmup=args.warmup,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
         device=local_rank if have_cuda else 0, print_json=False, converge=converge,
+        mode=args.mode,
     )
     barrier_sync()
     t0 = time.perf_counter()
@@ -308,6 +385,7 @@ def main():
         n_nodes=args.nodes, steps=args.steps, warmup=0,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
         device=local_rank if have_cuda else 0, print_json=False, converge=converge,
+        mode=args.mode,
     )
     barrier_sync()
     elapsed = time.perf_counter() - t0
@@ -348,6 +426,7 @@ def main():
                 "gpu_pods_per_node": 2,
                 "gpu_validation": gpu_validate,
                 "converging_reconcile": converge,
+                "mode": args.mode,
                 "global_batch": args.nodes * int(os.environ.get("WORLD_SIZE", "1")),
                 "parallelism": f"dp{world} (one independent simulated "
                                f"{args.nodes}-node cluster per rank)",

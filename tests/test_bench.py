@@ -35,3 +35,13 @@ def test_bench_cli_prints_contract_json():
     assert data["higher_is_better"] is False
     assert data["scaling"] == "weak"
     assert data["config"]["model"] == "amdgpu-driver-rolling-upgrade"
+
+
+def test_bench_requestor_mode():
+    import bench
+
+    result = bench.run_rolling_upgrade_benchmark(
+        n_nodes=2, steps=1, warmup=0, max_parallel=1, gpu_validate=False,
+        print_json=False, mode="requestor",
+    )
+    assert result["upgrades_completed"] == 1
