@@ -86,11 +86,18 @@ class Counter:
         self.name = name
         self.help = help_
         self._values: Dict[Tuple[str, ...], int] = {}
+        self._total = 0
         self._lock = threading.Lock()
 
     def inc(self, *labels: str, amount: int = 1) -> None:
         with self._lock:
             self._values[labels] = self._values.get(labels, 0) + amount
+            self._total += amount
+
+    def total(self) -> int:
+        """O(1) sum over all label sets."""
+        with self._lock:
+            return self._total
 
     def value(self, *labels: str) -> int:
         with self._lock:

@@ -75,6 +75,11 @@ class PodManager:
         self._nodes_in_progress = util.StringSet()
         self._workers: List[threading.Thread] = []
         self._workers_lock = threading.Lock()
+        # (namespace, name, resourceVersion) -> latest revision hash; a DS's
+        # hash can only change when the DS object itself changes, so this
+        # collapses the per-node ControllerRevision LISTs of a reconcile pass
+        # into one
+        self._ds_hash_cache: dict = {}
 
     # -- revision-hash helpers (pod_manager.go:84-118) -----------------------
 
@@ -90,7 +95,13 @@ class PodManager:
     def get_daemonset_controller_revision_hash(self, daemonset: K8sObject) -> str:
         """Latest ControllerRevision hash for the DaemonSet
         (pod_manager.go:92-118): list revisions by the DS's selector labels,
-        take the highest ``revision``, strip the ``<dsname>-`` name prefix."""
+        take the highest ``revision``, strip the ``<dsname>-`` name prefix.
+        Cached per DS resourceVersion."""
+        cache_key = (meta.namespace(daemonset), meta.name(daemonset),
+                     meta.resource_version(daemonset))
+        cached = self._ds_hash_cache.get(cache_key)
+        if cached is not None:
+            return cached
         selector = daemonset.get("spec", {}).get("selector", {}).get("matchLabels", {})
         label_selector = ",".join(f"{k}={v}" for k, v in sorted(selector.items()))
         revisions = self._client.list_controller_revisions(
@@ -103,7 +114,11 @@ class PodManager:
         latest = max(revisions, key=lambda r: r.get("revision", 0))
         prefix = f"{meta.name(daemonset)}-"
         name = meta.name(latest)
-        return name[len(prefix):] if name.startswith(prefix) else name
+        hash_ = name[len(prefix):] if name.startswith(prefix) else name
+        if len(self._ds_hash_cache) > 256:
+            self._ds_hash_cache.clear()
+        self._ds_hash_cache[cache_key] = hash_
+        return hash_
 
     # -- eviction (pod_manager.go:122-232) -----------------------------------
 

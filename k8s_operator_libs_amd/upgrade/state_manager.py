@@ -190,10 +190,13 @@ class ClusterUpgradeStateManager:
         )
 
         state_key = util.get_upgrade_state_label_key()
+        transitions = self.metrics.state_transitions
 
         def step(fn, *args):
+            before = transitions.total() if live else 0
             fn(*args)
-            if live:
+            # regroup only when the phase actually moved a node (most don't)
+            if live and transitions.total() != before:
                 current_state.regroup(state_key)
 
         common = self.common
