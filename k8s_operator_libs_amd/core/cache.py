@@ -49,16 +49,31 @@ class _Informer:
         self._thread = threading.Thread(target=self._run, daemon=True)
         self._thread.start()
 
-    def _run(self) -> None:
-        # initial LIST (after the watch opened, so no events are lost)
+    def _relist(self) -> None:
+        """Full LIST replacing the store (initial sync and watch-reconnect
+        recovery — a dropped stream may have lost events)."""
         objs = self._delegate.list(self.api_version, self.kind)
         with self._lock:
-            for obj in objs:
-                self._store[(meta.namespace(obj), meta.name(obj))] = obj
+            self._store = {
+                (meta.namespace(obj), meta.name(obj)): obj for obj in objs
+            }
+
+    def _run(self) -> None:
+        # initial LIST (after the watch opened, so no events are lost)
+        self._relist()
         self._synced.set()
         while not self._stop.is_set():
             item = self._watch.next(timeout=0.2)
             if item is None:
+                alive = getattr(self._watch, "alive", None)
+                if alive is not None and not alive():
+                    # stream dropped: reconnect, then relist to recover any
+                    # events lost in the gap (informer resync semantics)
+                    try:
+                        self._watch = self._delegate.watch(self.api_version, self.kind)
+                        self._relist()
+                    except Exception:
+                        self._stop.wait(1.0)
                 continue
             if self._sync_delay:
                 time.sleep(self._sync_delay)

@@ -68,6 +68,9 @@ class Watch:
         self._stopped = True
         self._cluster._remove_watch(self)
 
+    def alive(self) -> bool:
+        return not self._stopped
+
 
 class FakeCluster:
     """Thread-safe in-memory Kubernetes object store."""

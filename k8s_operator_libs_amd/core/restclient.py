@@ -296,3 +296,7 @@ class _HttpWatch:
 
     def stop(self) -> None:
         self._stop.set()
+
+    def alive(self) -> bool:
+        """False once the HTTP stream has ended (server drop or stop)."""
+        return self._thread.is_alive() and not self._stop.is_set()

@@ -101,3 +101,25 @@ def test_full_upgrade_over_lagging_cache(client):
             assert count == 1, f"transition {frm}->{to} fired {count} times"
     finally:
         cached.stop()
+
+
+def test_informer_reconnects_after_watch_drop(client):
+    """A dropped watch stream must not silently freeze the cache: the
+    informer reconnects and relists (resync semantics)."""
+    cached = CachedClient(client)
+    try:
+        NodeBuilder("n1").build(client.cluster)
+        cached.get_node("n1")  # start informer
+        inf = cached._informers[("v1", "Node")]
+        # simulate a server-side stream drop
+        inf._watch.stop()
+        # mutate while the stream is down
+        client.patch("v1", "Node", "n1", {"metadata": {"labels": {"x": "1"}}})
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            if cached.get_node("n1")["metadata"].get("labels", {}).get("x") == "1":
+                break
+            time.sleep(0.02)
+        assert cached.get_node("n1")["metadata"]["labels"]["x"] == "1"
+    finally:
+        cached.stop()
