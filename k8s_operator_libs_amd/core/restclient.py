@@ -62,6 +62,7 @@ class RestClient(Client):
         verify: Any = True,
         timeout: float = 30.0,
         extra_kinds: Optional[Dict[tuple, tuple]] = None,
+        retries: int = 3,
     ) -> None:
         self.base_url = base_url.rstrip("/")
         headers = {"Content-Type": "application/json"}
@@ -73,6 +74,30 @@ class RestClient(Client):
         self._kinds = dict(_KIND_INFO)
         if extra_kinds:
             self._kinds.update(extra_kinds)
+        self._retries = retries
+
+    def _request(self, method: str, path: str, **kw) -> httpx.Response:
+        """Issue a request with client-go-style retries on transient failures
+        (connection errors, 429/5xx).  Conflicts/NotFound/etc. surface
+        immediately — the callers' optimistic-concurrency logic owns those."""
+        import time as _time
+
+        attempt = 0
+        while True:
+            try:
+                resp = self._http.request(method, path, **kw)
+            except (httpx.ConnectError, httpx.ReadError, httpx.RemoteProtocolError,
+                    httpx.ConnectTimeout, httpx.ReadTimeout) as exc:
+                if attempt >= self._retries:
+                    err = ApiError(f"connection to apiserver failed: {exc}")
+                    err.code = 503
+                    raise err from exc
+            else:
+                if resp.status_code not in (429, 500, 502, 503, 504) or \
+                        attempt >= self._retries:
+                    return resp
+            attempt += 1
+            _time.sleep(min(0.05 * (2 ** attempt), 1.0))
 
     @classmethod
     def from_environment(cls) -> "RestClient":
@@ -160,7 +185,7 @@ class RestClient(Client):
     # -- Client implementation -------------------------------------------------
 
     def get(self, api_version, kind, name, namespace=""):
-        resp = self._http.get(self._object_path(api_version, kind, name, namespace))
+        resp = self._request("GET", self._object_path(api_version, kind, name, namespace))
         self._raise_for(resp)
         return resp.json()
 
@@ -171,35 +196,35 @@ class RestClient(Client):
         if field_selector:
             params["fieldSelector"] = field_selector
         path = self._collection_path(api_version, kind, namespace or "")
-        resp = self._http.get(path, params=params)
+        resp = self._request("GET", path, params=params)
         self._raise_for(resp)
         return resp.json().get("items", [])
 
     def create(self, obj):
         api_version, kind = meta.api_version(obj), meta.kind(obj)
         path = self._collection_path(api_version, kind, meta.namespace(obj))
-        resp = self._http.post(path, content=json.dumps(obj))
+        resp = self._request("POST", path, content=json.dumps(obj))
         self._raise_for(resp)
         return resp.json()
 
     def update(self, obj):
         api_version, kind = meta.api_version(obj), meta.kind(obj)
         path = self._object_path(api_version, kind, meta.name(obj), meta.namespace(obj))
-        resp = self._http.put(path, content=json.dumps(obj))
+        resp = self._request("PUT", path, content=json.dumps(obj))
         self._raise_for(resp)
         return resp.json()
 
     def patch(self, api_version, kind, name, patch, namespace=""):
         path = self._object_path(api_version, kind, name, namespace)
-        resp = self._http.patch(
-            path, content=json.dumps(patch),
+        resp = self._request(
+            "PATCH", path, content=json.dumps(patch),
             headers={"Content-Type": "application/merge-patch+json"},
         )
         self._raise_for(resp)
         return resp.json()
 
     def delete(self, api_version, kind, name, namespace=""):
-        resp = self._http.delete(self._object_path(api_version, kind, name, namespace))
+        resp = self._request("DELETE", self._object_path(api_version, kind, name, namespace))
         self._raise_for(resp)
 
     def evict_pod(self, name, namespace):
@@ -209,7 +234,7 @@ class RestClient(Client):
             "kind": "Eviction",
             "metadata": {"name": name, "namespace": namespace},
         }
-        resp = self._http.post(path, content=json.dumps(body))
+        resp = self._request("POST", path, content=json.dumps(body))
         self._raise_for(resp)
 
     def watch(self, api_version: str, kind: str):
@@ -223,7 +248,7 @@ class RestClient(Client):
 
     def discover_resource(self, api_version: str, plural: str) -> bool:
         prefix = f"/api/{api_version}" if "/" not in api_version else f"/apis/{api_version}"
-        resp = self._http.get(prefix)
+        resp = self._request("GET", prefix)
         if resp.status_code != 200:
             return False
         for res in resp.json().get("resources", []):
