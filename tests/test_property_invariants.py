@@ -1,0 +1,96 @@
+"""Property-based invariants (hypothesis) for the rolling-upgrade window.
+
+For arbitrary cluster sizes and policies, starting from a healthy all-Ready
+cluster, the state machine must:
+
+- complete every node to upgrade-done,
+- never exceed maxParallelUpgrades nodes in progress at any observation,
+- never exceed the maxUnavailable budget of cordoned nodes,
+- fire every (from,to) label transition at most once per node lifecycle.
+"""
+
+import math
+
+from hypothesis import given, settings, strategies as st
+
+from k8s_operator_libs_amd.api.upgrade.v1alpha1 import IntOrString
+from k8s_operator_libs_amd.core import FakeClient
+from k8s_operator_libs_amd.metrics import MetricsRegistry
+from k8s_operator_libs_amd.upgrade import consts, util
+from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
+
+from simenv import SimDaemonSetController
+from test_state_manager import policy, setup_cluster
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    n_nodes=st.integers(min_value=1, max_value=8),
+    max_parallel=st.integers(min_value=0, max_value=4),
+    max_unavailable_pct=st.sampled_from([25, 50, 100]),
+    drain=st.booleans(),
+    live=st.booleans(),
+)
+def test_rolling_window_invariants(n_nodes, max_parallel, max_unavailable_pct,
+                                   drain, live):
+    client = FakeClient()
+    reg = MetricsRegistry()
+    ds, _ = setup_cluster(client, n_nodes=n_nodes, pod_hash="old", ds_hash="new")
+    SimDaemonSetController(client.cluster, ds, current_hash="new")
+    manager = ClusterUpgradeStateManager(client, metrics=reg)
+    pol = policy(
+        maxParallelUpgrades=max_parallel,
+        maxUnavailable=f"{max_unavailable_pct}%",
+        drainSpec={"enable": drain},
+    )
+    state_key = util.get_upgrade_state_label_key()
+    budget = IntOrString.scaled_value(f"{max_unavailable_pct}%", n_nodes, True)
+
+    in_progress_states = set(consts.ALL_STATES) - {
+        consts.UPGRADE_STATE_UNKNOWN,
+        consts.UPGRADE_STATE_DONE,
+        consts.UPGRADE_STATE_UPGRADE_REQUIRED,
+    }
+
+    for tick in range(40 * n_nodes):
+        state = manager.reconcile(
+            "amd-gpu-operator", {"app": "amdgpu-driver-daemonset"}, pol,
+            converge=live,
+        )
+        nodes = client.list_nodes()
+        states = [n["metadata"]["labels"].get(state_key, "") for n in nodes]
+        in_progress = sum(s in in_progress_states for s in states)
+        cordoned = sum(bool(n["spec"].get("unschedulable")) for n in nodes)
+        if max_parallel > 0:
+            assert in_progress <= max_parallel, (
+                f"window exceeded: {in_progress} > {max_parallel} at tick {tick}"
+            )
+        assert cordoned <= max(budget, 1), (
+            f"unavailability budget exceeded: {cordoned} cordoned > {budget}"
+        )
+        for s in states:
+            assert s in consts.ALL_STATES
+        if all(s == consts.UPGRADE_STATE_DONE for s in states):
+            break
+    assert all(
+        n["metadata"]["labels"].get(state_key) == consts.UPGRADE_STATE_DONE
+        for n in client.list_nodes()
+    ), f"did not converge: {states}"
+    # every node uncordoned at the end
+    assert not any(n["spec"].get("unschedulable") for n in client.list_nodes())
+    # each transition fired exactly n_nodes times at most (once per node)
+    for (frm, to), count in reg.state_transitions.items().items():
+        assert count <= n_nodes, f"{frm}->{to} fired {count} times for {n_nodes} nodes"
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    total=st.integers(min_value=0, max_value=200),
+    pct=st.integers(min_value=0, max_value=100),
+)
+def test_int_or_percent_bounds(total, pct):
+    up = IntOrString.scaled_value(f"{pct}%", total, True)
+    down = IntOrString.scaled_value(f"{pct}%", total, False)
+    assert down <= up <= down + 1
+    assert up == math.ceil(pct * total / 100)
+    assert 0 <= up <= total or pct > 100
