@@ -167,6 +167,7 @@ def main(argv=None) -> int:
     parser.add_argument("--metrics-port", type=int, default=8080)
     parser.add_argument("--demo", action="store_true")
     parser.add_argument("--demo-nodes", type=int, default=8)
+    parser.add_argument("--no-leader-election", action="store_true")
     args = parser.parse_args(argv)
 
     logging.basicConfig(level=logging.INFO,
@@ -194,8 +195,21 @@ def main(argv=None) -> int:
         "podDeletion": {"deleteEmptyDir": True},
         "drainSpec": {"enable": True},
     })
-    reconcile_loop(manager, client, policy, args.namespace, driver_labels,
-                   interval=args.interval)
+
+    from k8s_operator_libs_amd.core.leaderelection import LeaderElector
+    from k8s_operator_libs_amd.upgrade.controller import UpgradeController
+
+    controller = UpgradeController(
+        manager, args.namespace, driver_labels, policy,
+        resync_seconds=max(args.interval, 1.0),
+    )
+    if args.no_leader_election:
+        controller.run()
+    else:
+        elector = LeaderElector(client, "amd-gpu-operator-upgrade",
+                                namespace=args.namespace)
+        elector.run(on_started_leading=controller.run,
+                    on_stopped_leading=controller.stop)
     return 0
 
 
