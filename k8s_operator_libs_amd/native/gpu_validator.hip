@@ -26,6 +26,7 @@
 
 #include <cmath>
 #include <algorithm>
+#include <cstdint>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -97,6 +98,98 @@ __global__ void mfma_bf16_16x16x32_kernel(const __bf16* __restrict__ A,
     D[row * 16 + col] = c[r];
   }
 #endif
+}
+
+// ---------------------------------------------------------------------------
+// MFMA fp8 (OCP e4m3fn) smoke: v_mfma_f32_16x16x32_fp8_fp8, one wave.
+// Same fragment geometry as the bf16 K=32 shape (8 elements per lane,
+// packed into one i64 operand; C/D layout is dtype-independent).
+// Validates the fp8 matrix path the serving stack depends on.
+// ---------------------------------------------------------------------------
+
+__global__ void mfma_fp8_16x16x32_kernel(const uint8_t* __restrict__ A,
+                                         const uint8_t* __restrict__ B,
+                                         float* __restrict__ D) {
+#if defined(__gfx950__)
+  int lane = threadIdx.x;
+  uint64_t a = 0, b = 0;
+  for (int i = 0; i < 8; ++i) {
+    int k = (lane >> 4) * 8 + i;
+    a |= (uint64_t)A[(lane & 15) * 32 + k] << (8 * i);  // A is 16x32
+    b |= (uint64_t)B[k * 16 + (lane & 15)] << (8 * i);  // B is 32x16
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8((long)a, (long)b, c, 0, 0, 0);
+  for (int r = 0; r < 4; ++r) {
+    D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = c[r];
+  }
+#endif
+}
+
+static float decode_e4m3fn(uint8_t u) {
+  int s = (u >> 7) & 1;
+  int e = (u >> 3) & 0xF;
+  int m = u & 7;
+  float v;
+  if (e == 0) {
+    v = (float)m / 8.0f * (1.0f / 64.0f);  // subnormal: 2^-6 scale
+  } else if (e == 15 && m == 7) {
+    v = 0.0f;  // NaN code; never produced by our encoder
+  } else {
+    v = (1.0f + (float)m / 8.0f) * std::ldexp(1.0f, e - 7);
+  }
+  return s ? -v : v;
+}
+
+static uint8_t encode_e4m3fn_nearest(float f) {
+  // exact nearest-representable search over the 256-code table (fine for a
+  // smoke test; avoids re-deriving RNE tie rules)
+  uint8_t best = 0;
+  float best_err = 1e30f;
+  for (int u = 0; u < 256; ++u) {
+    if ((u & 0x7F) == 0x7F) continue;  // NaN codes
+    float err = std::fabs(decode_e4m3fn((uint8_t)u) - f);
+    if (err < best_err) {
+      best_err = err;
+      best = (uint8_t)u;
+    }
+  }
+  return best;
+}
+
+static double mfma_fp8_check(int device) {
+  HIP_CHECK(hipSetDevice(device));
+  const int M = 16, N = 16, K = 32;
+  std::vector<uint8_t> hA(M * K), hB(K * N);
+  std::vector<float> hD(M * N), ref(M * N);
+  for (int i = 0; i < M * K; ++i)
+    hA[i] = encode_e4m3fn_nearest(0.05f * (float)((i * 5) % 23) - 0.5f);
+  for (int i = 0; i < K * N; ++i)
+    hB[i] = encode_e4m3fn_nearest(0.03f * (float)((i * 7) % 19) - 0.25f);
+  for (int m = 0; m < M; ++m)
+    for (int n = 0; n < N; ++n) {
+      float acc = 0.f;
+      for (int k = 0; k < K; ++k)
+        acc = fmaf(decode_e4m3fn(hA[m * K + k]), decode_e4m3fn(hB[k * N + n]), acc);
+      ref[m * N + n] = acc;
+    }
+  uint8_t *dA, *dB;
+  float* dD;
+  HIP_CHECK(hipMalloc(&dA, M * K));
+  HIP_CHECK(hipMalloc(&dB, K * N));
+  HIP_CHECK(hipMalloc(&dD, sizeof(float) * M * N));
+  HIP_CHECK(hipMemcpy(dA, hA.data(), M * K, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(dB, hB.data(), K * N, hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(mfma_fp8_16x16x32_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dD);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipMemcpy(hD.data(), dD, sizeof(float) * M * N, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(dA));
+  HIP_CHECK(hipFree(dB));
+  HIP_CHECK(hipFree(dD));
+  double max_err = 0.0;
+  for (int i = 0; i < M * N; ++i)
+    max_err = std::max(max_err, (double)std::fabs(hD[i] - ref[i]));
+  return max_err;
 }
 
 // ---------------------------------------------------------------------------
@@ -445,6 +538,9 @@ PYBIND11_MODULE(_gpu_validator, m) {
         py::arg("buf_mib") = 1024.0, py::arg("iters") = 10,
         "Streaming float4 copy bandwidth in GB/s (read+write)");
   m.def("lds_roundtrip_check", &lds_roundtrip_check, py::arg("device") = 0);
+  m.def("mfma_fp8_check", &mfma_fp8_check, py::arg("device") = 0,
+        "Max abs error of a v_mfma_f32_16x16x32_fp8_fp8 (OCP e4m3) tile vs "
+        "f32 CPU reference");
   m.def("mfma_throughput_tflops", &mfma_throughput_tflops, py::arg("device") = 0,
         py::arg("iters") = 200000,
         "Sustained bf16 matrix-core throughput in TFLOP/s (burn-in check)");

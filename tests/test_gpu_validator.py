@@ -77,3 +77,10 @@ def test_mfma_throughput_burn_in(native):
     # comfortably. Catches down-clocked / power-capped parts.
     tf = native.mfma_throughput_tflops(0, 100000)
     assert tf > 1000.0, f"matrix-core throughput suspiciously low: {tf} TF"
+
+
+def test_mfma_fp8_e4m3_close(native):
+    # OCP e4m3 inputs, f32 accumulate; f32 CPU reference over the quantized
+    # operands. Catches fp8-unit faults the bf16 path can't see.
+    err = native.mfma_fp8_check(0)
+    assert err <= 5e-2, f"fp8 MFMA error too large: {err}"
