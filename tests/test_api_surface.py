@@ -153,3 +153,14 @@ def test_event_recorder_writes_events(client):
     assert events[0]["reason"] == "AMDGPUDriverUpgrade"
     assert "upgrade-done" in events[0]["message"]
     assert events[0]["involvedObject"]["name"] == "n1"
+
+
+def test_log_level_mapping():
+    import logging
+
+    from k8s_operator_libs_amd import consts
+
+    assert consts.to_logging_level(consts.LOG_LEVEL_ERROR) == logging.ERROR
+    assert consts.to_logging_level(consts.LOG_LEVEL_INFO) == logging.INFO
+    assert consts.to_logging_level(5) == logging.DEBUG
+    assert consts.to_logging_level(-9) == logging.ERROR

@@ -168,3 +168,23 @@ def test_apply_crds_cli_example(tmp_path, crd_dir, capsys):
 
     rc = cli.main(["--crds-path", crd_dir, "--operation", "apply", "--fake"])
     assert rc == 0
+
+
+def test_generated_consumer_crd_applies(client, tmp_path):
+    """examples/generate_crd.py output is a valid, servable CRD whose policy
+    schema round-trips through crdutil apply."""
+    import examples.generate_crd as gen
+    import yaml as _yaml
+
+    crd_path = tmp_path / "amdgpudrivers.yaml"
+    crd_path.write_text(_yaml.safe_dump(gen.build_crd(), sort_keys=False))
+    assert process_crds(client, [str(crd_path)], CRD_OPERATION_APPLY) == 1
+    client.create({
+        "apiVersion": "driver.amd.com/v1alpha1", "kind": "AMDGPUDriver",
+        "metadata": {"name": "default"},
+        "spec": {"driverVersion": "6.4",
+                 "driverUpgradePolicy": {"autoUpgrade": True,
+                                          "maxParallelUpgrades": 2}},
+    })
+    got = client.get("driver.amd.com/v1alpha1", "AMDGPUDriver", "default")
+    assert got["spec"]["driverUpgradePolicy"]["maxParallelUpgrades"] == 2
