@@ -12,9 +12,11 @@
 //                           chain (matrix cores, exact f32 numerics)
 //   mfma_bf16_check()     — v_mfma_f32_16x16x32_bf16 tile vs CPU reference
 //                           (the production-dtype matrix path)
-//   hbm_bandwidth_gbps()  — float4 streaming copy, grid sized to cover all
-//                           256 CUs across the 8 XCDs
+//   mfma_fp8_check()      — v_mfma_f32_16x16x32_fp8_fp8 (OCP e4m3) tile
+//   mfma_throughput_tflops() — sustained bf16 matrix-core burn-in
+//   hbm_bandwidth_gbps()  — nontemporal streaming copy (sweep-tuned, ~6 TB/s)
 //   lds_roundtrip_check() — LDS store/load/barrier integrity
+//   xgmi_p2p_probe()      — peer reachability + p2p bandwidth per xGMI link
 //
 // Built standalone with hipcc (no torch linkage) via pybind11; the .so lives
 // in-tree so it travels to GPU nodes with the package.

@@ -12,8 +12,11 @@ from typing import Any, Dict
 # Acceptance thresholds for an MI355X node.
 MFMA_F32_MAX_ERR = 1e-6       # exact fmaf-chain numerics: effectively zero
 MFMA_BF16_MAX_ERR = 5e-2      # bf16 inputs, f32 accumulate, K=32
-HBM_MIN_GBPS = 1000.0         # far below the ~6300 GB/s measured copy BW,
+MFMA_FP8_MAX_ERR = 5e-2       # OCP e4m3 inputs, f32 accumulate
+HBM_MIN_GBPS = 1000.0         # far below the ~6000 GB/s measured copy BW,
                               # catches catastrophically degraded memory
+MFMA_MIN_TFLOPS = 1000.0      # deep mode: half the bf16 issue-rate floor,
+                              # catches down-clocked / power-capped parts
 
 
 class GpuHealthError(RuntimeError):
@@ -83,6 +86,7 @@ def gpu_health_check(
     bw_buf_mib: float = 512.0,
     bw_iters: int = 5,
     require_gpu: bool = False,
+    deep: bool = False,
 ) -> Dict[str, Any]:
     """Full node GPU health check.  Returns a report dict with a top-level
     ``healthy`` verdict; raises GpuHealthError when ``require_gpu`` and no
@@ -113,6 +117,15 @@ def gpu_health_check(
     lds_ok = native.lds_roundtrip_check(device)
     report["checks"]["lds_ok"] = lds_ok
 
+    deep_ok = True
+    if deep:
+        # deep mode: fp8 matrix path + sustained matrix-core burn-in
+        fp8_err = native.mfma_fp8_check(device)
+        tflops = native.mfma_throughput_tflops(device, 100000)
+        report["checks"]["mfma_fp8_max_err"] = fp8_err
+        report["checks"]["mfma_throughput_tflops"] = tflops
+        deep_ok = fp8_err <= MFMA_FP8_MAX_ERR and tflops >= MFMA_MIN_TFLOPS
+
     # multi-GPU nodes: verify every xGMI peer link is up
     xgmi_ok = True
     if probe.get("device_count", 1) > 1:
@@ -128,13 +141,17 @@ def gpu_health_check(
         and bw >= HBM_MIN_GBPS
         and lds_ok
         and xgmi_ok
+        and deep_ok
     )
     return report
 
 
 def main() -> int:
-    """CLI entry point for use inside a validation pod."""
-    report = gpu_health_check(require_gpu=True)
+    """CLI entry point for use inside a validation pod.  Pass ``--deep`` for
+    the fp8 + burn-in checks."""
+    import sys
+
+    report = gpu_health_check(require_gpu=True, deep="--deep" in sys.argv)
     print(json.dumps(report, indent=2, default=str))
     return 0 if report["healthy"] else 1
 

@@ -84,3 +84,13 @@ def test_mfma_fp8_e4m3_close(native):
     # operands. Catches fp8-unit faults the bf16 path can't see.
     err = native.mfma_fp8_check(0)
     assert err <= 5e-2, f"fp8 MFMA error too large: {err}"
+
+
+def test_deep_health_check_on_gpu():
+    from k8s_operator_libs_amd.validation import gpu_health_check
+
+    report = gpu_health_check(require_gpu=True, deep=True,
+                              bw_buf_mib=256.0, bw_iters=3)
+    assert report["healthy"], report
+    assert report["checks"]["mfma_fp8_max_err"] <= 5e-2
+    assert report["checks"]["mfma_throughput_tflops"] > 1000.0
