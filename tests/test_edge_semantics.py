@@ -140,3 +140,61 @@ class TestMultiDriverDaemonSets:
         assert len(entries) == 2
         # both entries share the same node object (fetch dedup)
         assert entries[0].node is entries[1].node
+
+
+class TestSafeLoadUnblockPoints:
+    def test_validation_phase_unblocks_safe_load(self, client):
+        """The driver may restart after reaching validation-required and
+        block on safe load again; the validation phase must unblock it
+        (common_manager.go:581-586)."""
+        manager = ClusterUpgradeStateManager(client).with_validation_enabled(
+            "app=amd-gpu-validator"
+        )
+        setup_cluster(client, node_states=consts.UPGRADE_STATE_VALIDATION_REQUIRED)
+        key = util.get_upgrade_wait_for_safe_driver_load_annotation_key()
+        client.patch("v1", "Node", "node-0",
+                     {"metadata": {"annotations": {key: "true"}}})
+        # no validator pod yet -> validation does not pass, but the safe-load
+        # annotation must already be gone after the phase runs
+        state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+        manager.apply_state(state, policy())
+        assert key not in client.get_node("node-0")["metadata"]["annotations"]
+        assert state_of(client, "node-0") == consts.UPGRADE_STATE_VALIDATION_REQUIRED
+
+
+class TestInitContainerCrashLoop:
+    def test_failing_init_container_fails_upgrade(self, client):
+        """isDriverPodFailing also inspects initContainerStatuses
+        (common_manager.go:636-648)."""
+        manager = ClusterUpgradeStateManager(client)
+        ds, _ = setup_cluster(client, node_states=consts.UPGRADE_STATE_POD_RESTART_REQUIRED,
+                              pod_ready=False)
+        pod = client.list_pods(namespace=DRIVER_NS)[0]
+        client.patch("v1", "Pod", pod["metadata"]["name"],
+                     {"status": {
+                         "containerStatuses": [{"name": "driver", "ready": False,
+                                                "restartCount": 0}],
+                         "initContainerStatuses": [{"name": "safe-load-gate",
+                                                    "ready": False,
+                                                    "restartCount": 11}]}},
+                     DRIVER_NS)
+        state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+        manager.apply_state(state, policy())
+        assert state_of(client, "node-0") == consts.UPGRADE_STATE_FAILED
+
+    def test_ready_init_container_with_restarts_is_fine(self, client):
+        manager = ClusterUpgradeStateManager(client)
+        setup_cluster(client, node_states=consts.UPGRADE_STATE_POD_RESTART_REQUIRED,
+                      pod_ready=False)
+        pod = client.list_pods(namespace=DRIVER_NS)[0]
+        client.patch("v1", "Pod", pod["metadata"]["name"],
+                     {"status": {
+                         "containerStatuses": [{"name": "driver", "ready": False,
+                                                "restartCount": 2}],
+                         "initContainerStatuses": [{"name": "g", "ready": True,
+                                                    "restartCount": 50}]}},
+                     DRIVER_NS)
+        state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
+        manager.apply_state(state, policy())
+        # ready init container with many restarts doesn't count as failing
+        assert state_of(client, "node-0") == consts.UPGRADE_STATE_POD_RESTART_REQUIRED
