@@ -218,3 +218,33 @@ class TestMetaHelpers:
         t = {"a": {"b": 1, "c": 2}, "d": 3}
         json_merge_patch(t, {"a": {"b": None, "e": 4}, "d": 5})
         assert t == {"a": {"c": 2, "e": 4}, "d": 5}
+
+
+class TestRfc7386Vectors:
+    """The JSON merge patch test cases from RFC 7386 Appendix A."""
+
+    VECTORS = [
+        ({"a": "b"}, {"a": "c"}, {"a": "c"}),
+        ({"a": "b"}, {"b": "c"}, {"a": "b", "b": "c"}),
+        ({"a": "b"}, {"a": None}, {}),
+        ({"a": "b", "b": "c"}, {"a": None}, {"b": "c"}),
+        ({"a": ["b"]}, {"a": "c"}, {"a": "c"}),
+        ({"a": "c"}, {"a": ["b"]}, {"a": ["b"]}),
+        ({"a": {"b": "c"}}, {"a": {"b": "d", "c": None}}, {"a": {"b": "d"}}),
+        ({"a": [{"b": "c"}]}, {"a": [1]}, {"a": [1]}),
+        (["a", "b"], ["c", "d"], ["c", "d"]),
+        ({"a": "b"}, ["c"], ["c"]),
+        ({"a": "foo"}, None, None),
+        ({"a": "foo"}, "bar", "bar"),
+        ({"e": None}, {"a": 1}, {"e": None, "a": 1}),
+        ([1, 2], {"a": "b", "c": None}, {"a": "b"}),
+        ({}, {"a": {"bb": {"ccc": None}}}, {"a": {"bb": {}}}),
+    ]
+
+    def test_all_vectors(self):
+        for original, patch, expected in self.VECTORS:
+            result = json_merge_patch(
+                original if not isinstance(original, dict) else dict(original),
+                patch,
+            )
+            assert result == expected, (original, patch, result, expected)
