@@ -64,29 +64,6 @@ def serve_metrics(manager, port: int):
     return server
 
 
-def reconcile_loop(manager, client, policy, namespace, driver_labels,
-                   interval: float, max_iterations: int = 0):
-    state_key = util.get_upgrade_state_label_key()
-    i = 0
-    while True:
-        i += 1
-        try:
-            state = manager.build_state(namespace, driver_labels)
-            manager.apply_state(state, policy)
-        except Exception:
-            log.exception("reconcile failed; will retry")
-        else:
-            counts = manager.counts(state)
-            log.info("reconcile #%d: %s", i, counts)
-            if max_iterations and counts["total"] > 0 and \
-               counts["done"] == counts["total"]:
-                log.info("all %d nodes upgraded", counts["total"])
-                return True
-        if max_iterations and i >= max_iterations:
-            return False
-        time.sleep(interval)
-
-
 def run_demo(args):
     """Self-contained demo: 8 synthetic MI355X nodes on the mini-apiserver."""
     import bench as bench_mod

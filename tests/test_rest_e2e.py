@@ -171,3 +171,34 @@ def test_cached_client_over_rest(rest, server):
         assert cached.get_node("cn1")["metadata"]["labels"]["x"] == "1"
     finally:
         cached.stop()
+
+
+def test_concurrent_rest_clients(rest, server):
+    """Thread-safety of the HTTP apiserver under concurrent writers."""
+    import threading
+
+    errors = []
+
+    def worker(i):
+        try:
+            c = RestClient(server.url)
+            for j in range(20):
+                name = f"cw-{i}-{j}"
+                c.create({"apiVersion": "v1", "kind": "Pod",
+                          "metadata": {"name": name, "namespace": "default"},
+                          "spec": {"nodeName": f"n{i}"}})
+                c.patch("v1", "Pod", name,
+                        {"metadata": {"labels": {"round": str(j)}}}, "default")
+            c.close()
+        except Exception as exc:
+            errors.append(exc)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errors, errors
+    assert len(rest.list_pods(namespace="default")) == 80
+    # per-node index stayed consistent under concurrency
+    assert len(rest.list_pods(field_selector="spec.nodeName=n2")) == 20
