@@ -155,6 +155,7 @@ class LeaderElector:
         """Block: campaign, then call ``on_started_leading()`` once leading
         (it should run until it observes lost leadership / stop), renewing in
         a background thread.  Returns when stopped."""
+        self._on_stopped_leading = on_stopped_leading
         while not self._stop.is_set():
             if self._try_acquire_or_renew():
                 logger.info("leader election: %s acquired %s/%s",
@@ -179,5 +180,14 @@ class LeaderElector:
                 logger.warning("leader election: %s lost %s/%s",
                                self.identity, self.namespace, self.lease_name)
                 self._leading.clear()
+                # Leadership lost while the leader's work is still running:
+                # fire the callback so the work is STOPPED (without this a
+                # demoted replica would keep reconciling — split-brain).
+                cb = getattr(self, "_on_stopped_leading", None)
+                if cb is not None:
+                    try:
+                        cb()
+                    except Exception:
+                        logger.exception("on_stopped_leading callback failed")
                 return
             self._stop.wait(self.retry_period)

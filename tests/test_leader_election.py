@@ -65,3 +65,34 @@ def test_second_candidate_waits_then_takes_over(client):
     b.stop()
     ta.join(timeout=5)
     tb.join(timeout=5)
+
+
+def test_lost_lease_stops_leader_work(client):
+    """When renewal fails mid-flight, the on_stopped_leading callback must
+    fire so the demoted replica stops reconciling (split-brain guard)."""
+    client.cluster.register_kind("coordination.k8s.io/v1", "Lease", "leases", True)
+    a = LeaderElector(client, "op", identity="a", lease_duration=0.4, retry_period=0.05)
+    working = threading.Event()
+    stopped = threading.Event()
+
+    def lead():
+        working.set()
+        while not stopped.is_set():
+            time.sleep(0.02)
+
+    t = threading.Thread(
+        target=lambda: a.run(lead, on_stopped_leading=stopped.set), daemon=True
+    )
+    t.start()
+    assert working.wait(5.0)
+    # usurper: another candidate force-takes the lease (e.g. after a clock
+    # hiccup or apiserver partition on a's side)
+    lease = client.get("coordination.k8s.io/v1", "Lease", "op", "default")
+    client.patch("coordination.k8s.io/v1", "Lease", "op",
+                 {"spec": {"holderIdentity": "b", "renewTime": lease["spec"]["renewTime"]}},
+                 "default")
+    # a's next renew sees a different holder with a fresh-enough lease... it
+    # must detect the loss and stop its work
+    assert stopped.wait(5.0), "leader work not stopped after losing the lease"
+    a.stop()
+    t.join(timeout=5)
