@@ -258,7 +258,15 @@ class PodManager:
         if key not in annotations:
             self._provider.change_node_upgrade_annotation(node, key, str(now))
             return
-        start_time = int(annotations[key])
+        try:
+            start_time = int(annotations[key])
+        except ValueError:
+            # corrupt stamp (manual edit?): re-stamp rather than wedging the
+            # phase forever (improves on the reference, which errors out)
+            logger.warning("node %s: corrupt completion start-time %r; re-stamping",
+                           meta.name(node), annotations[key])
+            self._provider.change_node_upgrade_annotation(node, key, str(now))
+            return
         if now > start_time + timeout_seconds:
             self._provider.change_node_upgrade_state(
                 node, consts.UPGRADE_STATE_POD_DELETION_REQUIRED

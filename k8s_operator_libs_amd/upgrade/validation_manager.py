@@ -97,7 +97,13 @@ class ValidationManager:
         if key not in annotations:
             self._provider.change_node_upgrade_annotation(node, key, str(now))
             return
-        start_time = int(annotations[key])
+        try:
+            start_time = int(annotations[key])
+        except ValueError:
+            logger.warning("node %s: corrupt validation start-time %r; re-stamping",
+                           meta.name(node), annotations[key])
+            self._provider.change_node_upgrade_annotation(node, key, str(now))
+            return
         if now > start_time + timeout_seconds:
             self._provider.change_node_upgrade_state(node, consts.UPGRADE_STATE_FAILED)
             logger.info("validation timeout exceeded on node %s -> upgrade-failed",

@@ -198,3 +198,52 @@ class TestInitContainerCrashLoop:
         manager.apply_state(state, policy())
         # ready init container with many restarts doesn't count as failing
         assert state_of(client, "node-0") == consts.UPGRADE_STATE_POD_RESTART_REQUIRED
+
+
+class TestCorruptTimeoutAnnotations:
+    def test_corrupt_completion_stamp_restamps(self, client):
+        from k8s_operator_libs_amd.api.upgrade.v1alpha1 import WaitForCompletionSpec
+        from k8s_operator_libs_amd.upgrade.node_state_provider import (
+            NodeUpgradeStateProvider,
+        )
+        from k8s_operator_libs_amd.upgrade.pod_manager import (
+            PodManager,
+            PodManagerConfig,
+        )
+
+        node = NodeBuilder("n1").with_upgrade_state(
+            consts.UPGRADE_STATE_WAIT_FOR_JOBS_REQUIRED
+        ).build(client.cluster)
+        PodBuilder("job", node="n1").with_labels({"app": "job"}).build(client.cluster)
+        key = util.get_wait_for_pod_completion_start_time_annotation_key()
+        client.patch("v1", "Node", "n1",
+                     {"metadata": {"annotations": {key: "garbage"}}})
+        node = client.get_node("n1")
+        mgr = PodManager(client, NodeUpgradeStateProvider(client))
+        mgr.schedule_check_on_pod_completion(PodManagerConfig(
+            nodes=[node],
+            wait_for_completion_spec=WaitForCompletionSpec(
+                podSelector="app=job", timeoutSecond=300),
+        ))
+        stamped = client.get_node("n1")["metadata"]["annotations"][key]
+        assert stamped.isdigit(), "corrupt stamp must be replaced, not crash"
+
+    def test_corrupt_validation_stamp_restamps(self, client):
+        from k8s_operator_libs_amd.upgrade.node_state_provider import (
+            NodeUpgradeStateProvider,
+        )
+        from k8s_operator_libs_amd.upgrade.validation_manager import ValidationManager
+
+        node = NodeBuilder("n1").build(client.cluster)
+        PodBuilder("val", node="n1").with_labels(
+            {"app": "amd-gpu-validator"}
+        ).not_ready().build(client.cluster)
+        key = util.get_validation_start_time_annotation_key()
+        client.patch("v1", "Node", "n1",
+                     {"metadata": {"annotations": {key: "not-a-number"}}})
+        node = client.get_node("n1")
+        mgr = ValidationManager(client, NodeUpgradeStateProvider(client),
+                                pod_selector="app=amd-gpu-validator")
+        assert mgr.validate(node) is False
+        stamped = client.get_node("n1")["metadata"]["annotations"][key]
+        assert stamped.isdigit()
