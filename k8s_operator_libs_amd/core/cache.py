@@ -59,8 +59,14 @@ class _Informer:
             }
 
     def _run(self) -> None:
-        # initial LIST (after the watch opened, so no events are lost)
-        self._relist()
+        # initial LIST (after the watch opened, so no events are lost);
+        # retried so a briefly-unavailable apiserver can't kill the informer
+        while not self._stop.is_set():
+            try:
+                self._relist()
+                break
+            except Exception:
+                self._stop.wait(0.5)
         self._synced.set()
         while not self._stop.is_set():
             item = self._watch.next(timeout=0.2)
