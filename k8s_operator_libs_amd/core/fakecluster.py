@@ -45,6 +45,7 @@ _BUILTIN_KINDS: Dict[Tuple[str, str], Tuple[str, bool]] = {
     # AMD maintenance-operator API (requestor mode; the reference's analogue
     # is maintenance.nvidia.com/v1alpha1 NodeMaintenance).
     ("maintenance.amd.com/v1alpha1", "NodeMaintenance"): ("nodemaintenances", True),
+    ("coordination.k8s.io/v1", "Lease"): ("leases", True),
 }
 
 

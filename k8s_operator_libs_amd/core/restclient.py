@@ -46,6 +46,7 @@ _KIND_INFO = {
     ("apiextensions.k8s.io/v1", "CustomResourceDefinition"): (
         "customresourcedefinitions", False),
     ("maintenance.amd.com/v1alpha1", "NodeMaintenance"): ("nodemaintenances", True),
+    ("coordination.k8s.io/v1", "Lease"): ("leases", True),
 }
 
 

@@ -7,7 +7,6 @@ from k8s_operator_libs_amd.core.leaderelection import LeaderElector
 
 
 def test_single_candidate_acquires_and_releases(client):
-    client.cluster.register_kind("coordination.k8s.io/v1", "Lease", "leases", True)
     elector = LeaderElector(client, "amd-gpu-operator", identity="a",
                             lease_duration=2.0, retry_period=0.05)
     ran = threading.Event()
@@ -30,7 +29,6 @@ def test_single_candidate_acquires_and_releases(client):
 
 
 def test_second_candidate_waits_then_takes_over(client):
-    client.cluster.register_kind("coordination.k8s.io/v1", "Lease", "leases", True)
     a = LeaderElector(client, "op", identity="a", lease_duration=0.5, retry_period=0.05)
     b = LeaderElector(client, "op", identity="b", lease_duration=0.5, retry_period=0.05)
     a_leading = threading.Event()
@@ -70,7 +68,6 @@ def test_second_candidate_waits_then_takes_over(client):
 def test_lost_lease_stops_leader_work(client):
     """When renewal fails mid-flight, the on_stopped_leading callback must
     fire so the demoted replica stops reconciling (split-brain guard)."""
-    client.cluster.register_kind("coordination.k8s.io/v1", "Lease", "leases", True)
     a = LeaderElector(client, "op", identity="a", lease_duration=0.4, retry_period=0.05)
     working = threading.Event()
     stopped = threading.Event()
@@ -96,3 +93,35 @@ def test_lost_lease_stops_leader_work(client):
     assert stopped.wait(5.0), "leader work not stopped after losing the lease"
     a.stop()
     t.join(timeout=5)
+
+
+def test_leader_election_over_rest():
+    """Lease CRUD works over the wire (mini-apiserver serves coordination.k8s.io)."""
+    import threading
+    import time as _time
+
+    from k8s_operator_libs_amd.core.apiserver import start_apiserver
+    from k8s_operator_libs_amd.core.restclient import RestClient
+
+    handle = start_apiserver()
+    rest = RestClient(handle.url)
+    try:
+        elector = LeaderElector(rest, "rest-op", identity="r1",
+                                lease_duration=2.0, retry_period=0.05)
+        led = threading.Event()
+
+        def lead():
+            led.set()
+            while elector.is_leading() and not elector._stop.is_set():
+                _time.sleep(0.02)
+
+        t = threading.Thread(target=lambda: elector.run(lead), daemon=True)
+        t.start()
+        assert led.wait(5.0)
+        lease = rest.get("coordination.k8s.io/v1", "Lease", "rest-op", "default")
+        assert lease["spec"]["holderIdentity"] == "r1"
+        elector.stop()
+        t.join(timeout=5)
+    finally:
+        rest.close()
+        handle.stop()
