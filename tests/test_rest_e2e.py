@@ -202,3 +202,35 @@ def test_concurrent_rest_clients(rest, server):
     assert len(rest.list_pods(namespace="default")) == 80
     # per-node index stayed consistent under concurrency
     assert len(rest.list_pods(field_selector="spec.nodeName=n2")) == 20
+
+
+def test_requestor_mode_over_rest(rest, server):
+    """Full requestor-mode lifecycle through HTTP: NodeMaintenance CRUD on
+    the named API group, shared finalizer flow, node completion."""
+    from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
+    from k8s_operator_libs_amd.upgrade.state_manager import StateOptions
+    from simenv import SimMaintenanceOperator
+
+    class W:
+        cluster = server.cluster
+
+    ds, _ = setup_cluster(W, pod_hash="old", ds_hash="new")
+    SimDaemonSetController(server.cluster, ds, current_hash="new")
+    SimMaintenanceOperator(server.cluster)
+    manager = ClusterUpgradeStateManager(
+        rest,
+        options=StateOptions(requestor=RequestorOptions(
+            use_maintenance_operator=True,
+            requestor_id="amd.gpu.operator",
+            namespace="default",
+        )),
+    )
+    pol = policy(drainSpec={"enable": True})
+    for _ in range(12):
+        manager.reconcile(DRIVER_NS, DRIVER_LABELS, pol)
+        if state_of(rest, "node-0") == consts.UPGRADE_STATE_DONE:
+            break
+    assert state_of(rest, "node-0") == consts.UPGRADE_STATE_DONE
+    with pytest.raises(NotFoundError):
+        rest.get("maintenance.amd.com/v1alpha1", "NodeMaintenance",
+                 "amd-operator-node-0", "default")
