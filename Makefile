@@ -5,7 +5,7 @@
 PY ?= python3
 HIPCC ?= /opt/rocm/bin/hipcc
 
-.PHONY: all build test test-gpu bench demo lint coverage clean
+.PHONY: all build test test-par test-gpu bench demo lint coverage clean
 
 all: build test
 
@@ -17,6 +17,11 @@ build:
 # HTTP mini apiservers).
 test:
 	$(PY) -m pytest tests/ -q -m "not gpu" --timeout 300
+
+# CPU suite under pytest-xdist (4 workers): shakes out cross-test
+# interference and shared-state races.
+test-par:
+	$(PY) -m pytest tests/ -q -m "not gpu" -n 4 --timeout 300
 
 # GPU test suite: requires an MI355X with ROCm.
 test-gpu:
