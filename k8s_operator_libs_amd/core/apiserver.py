@@ -207,6 +207,27 @@ def create_app(cluster: Optional[FakeCluster] = None):
     app.add_api_route("/apis/{group}/{version}/namespaces/{namespace}/{plural}/{name}",
                       _delete, methods=["DELETE"])
 
+    # status subresource: merge only the status field
+    async def _patch_status(group, version, plural, name, request: Request,
+                            namespace: str = ""):
+        body = json.loads(await request.body())
+
+        def run():
+            api_version, kind = _resolve(group, version, plural)
+            status_patch = {"status": body.get("status", body)}
+            return cluster.patch(api_version, kind, name, status_patch, namespace)
+        return _handle(run)
+
+    app.add_api_route("/api/{version}/{plural}/{name}/status",
+                      _wrap_nogroup_status(_patch_status), methods=["PATCH", "PUT"])
+    app.add_api_route("/api/{version}/namespaces/{namespace}/{plural}/{name}/status",
+                      _wrap_nogroup_status(_patch_status), methods=["PATCH", "PUT"])
+    app.add_api_route("/apis/{group}/{version}/{plural}/{name}/status",
+                      _patch_status, methods=["PATCH", "PUT"])
+    app.add_api_route(
+        "/apis/{group}/{version}/namespaces/{namespace}/{plural}/{name}/status",
+        _patch_status, methods=["PATCH", "PUT"])
+
     # pod eviction subresource
     @app.post("/api/v1/namespaces/{namespace}/pods/{name}/eviction")
     async def evict(namespace: str, name: str):
@@ -216,6 +237,13 @@ def create_app(cluster: Optional[FakeCluster] = None):
         return _handle(run)
 
     return app, cluster
+
+
+def _wrap_nogroup_status(fn):
+    async def handler(version: str, plural: str, name: str, request: Request,
+                      namespace: str = ""):
+        return await fn("", version, plural, name, request, namespace)
+    return handler
 
 
 def _wrap_nogroup(fn):

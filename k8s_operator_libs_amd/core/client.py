@@ -52,6 +52,15 @@ class Client(abc.ABC):
     def watch(self, api_version: str, kind: str) -> Watch:
         raise NotImplementedError("this client does not support watches")
 
+    def patch_status(
+        self, api_version: str, kind: str, name: str, status: dict, namespace: str = ""
+    ) -> K8sObject:
+        """Patch only the status subresource (controller-runtime
+        ``Status().Patch`` analogue).  Default implementation merges via the
+        main resource; clients talking to real apiservers override with the
+        ``/status`` endpoint."""
+        return self.patch(api_version, kind, name, {"status": status}, namespace)
+
     # -- typed conveniences used throughout pkg upgrade ----------------------
 
     def get_node(self, name: str) -> K8sObject:

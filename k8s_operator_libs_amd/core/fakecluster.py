@@ -15,6 +15,7 @@ which serves it over HTTP so the REST client can be tested wire-level.
 from __future__ import annotations
 
 import itertools
+import logging
 import queue
 import threading
 import time
@@ -143,7 +144,14 @@ class FakeCluster:
         for w in watches or ():
             w.events.put((event_type, snapshot))
         for hook in self._change_hooks:
-            hook(event_type, snapshot)
+            # hooks are observers (simulated controllers): a failing hook
+            # must never fail the API mutation that triggered it
+            try:
+                hook(event_type, snapshot)
+            except Exception:
+                logging.getLogger(__name__).exception(
+                    "change hook failed for %s %s", event_type, meta.name(obj)
+                )
 
     def _index_pod(self, obj: K8sObject, remove: bool = False) -> None:
         if meta.kind(obj) != "Pod":

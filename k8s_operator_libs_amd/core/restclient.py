@@ -228,6 +228,15 @@ class RestClient(Client):
         resp = self._request("DELETE", self._object_path(api_version, kind, name, namespace))
         self._raise_for(resp)
 
+    def patch_status(self, api_version, kind, name, status, namespace=""):
+        path = self._object_path(api_version, kind, name, namespace) + "/status"
+        resp = self._request(
+            "PATCH", path, content=json.dumps({"status": status}),
+            headers={"Content-Type": "application/merge-patch+json"},
+        )
+        self._raise_for(resp)
+        return resp.json()
+
     def evict_pod(self, name, namespace):
         path = self._object_path("v1", "Pod", name, namespace) + "/eviction"
         body = {

@@ -234,3 +234,17 @@ def test_requestor_mode_over_rest(rest, server):
     with pytest.raises(NotFoundError):
         rest.get("maintenance.amd.com/v1alpha1", "NodeMaintenance",
                  "amd-operator-node-0", "default")
+
+
+def test_status_subresource(rest, server):
+    rest.create({"apiVersion": "maintenance.amd.com/v1alpha1",
+                 "kind": "NodeMaintenance",
+                 "metadata": {"name": "nm1", "namespace": "default"},
+                 "spec": {"nodeName": "n1", "requestorID": "op"}})
+    rest.patch_status("maintenance.amd.com/v1alpha1", "NodeMaintenance", "nm1",
+                      {"conditions": [{"type": "Ready", "status": "True",
+                                       "reason": "Ready"}]}, "default")
+    live = rest.get("maintenance.amd.com/v1alpha1", "NodeMaintenance", "nm1", "default")
+    assert live["status"]["conditions"][0]["reason"] == "Ready"
+    # spec untouched by the status patch
+    assert live["spec"]["requestorID"] == "op"
