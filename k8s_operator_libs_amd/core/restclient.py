@@ -308,9 +308,15 @@ class _HttpWatch:
                             self._connected.set()
                             continue
                         self._queue.put((event.get("type"), event.get("object")))
-            except Exception:
+            except Exception as exc:
                 if not self._stop.is_set():
-                    raise
+                    # stream dropped: alive() goes False and the informer
+                    # layer reconnects + relists; no point crashing a thread
+                    import logging as _logging
+
+                    _logging.getLogger(__name__).warning(
+                        "watch stream ended: %s", exc
+                    )
             finally:
                 self._connected.set()  # never leave a waiter hanging
 
