@@ -201,7 +201,9 @@ class FakeCluster:
             bucket = self._bucket(api_version, kind)
             if (ns, name_) in bucket:
                 raise AlreadyExistsError(f"{kind} {ns}/{name_} already exists")
-            md["uid"] = str(uuid.uuid4())
+            # respect a caller-provided uid (snapshot load, fixtures with
+            # pre-wired ownerReferences); assign one otherwise
+            md["uid"] = md.get("uid") or str(uuid.uuid4())
             md["resourceVersion"] = self._next_rv()
             md.setdefault("creationTimestamp", _now_iso())
             md.setdefault("generation", 1)
@@ -385,6 +387,35 @@ class FakeCluster:
         crd.setdefault("status", {})["conditions"] = [
             {"type": "Established", "status": "True", "reason": "InitialNamesAccepted"}
         ]
+
+    # -- snapshot dump/load (debugging, deterministic replays) ----------------
+
+    def dump(self) -> dict:
+        """Serializable snapshot of every stored object (plus the dynamic
+        kind registry), for bug reports and replay tests."""
+        with self._lock:
+            return {
+                "kinds": [
+                    {"apiVersion": av, "kind": k, "plural": pl, "namespaced": ns}
+                    for (av, k), (pl, ns) in sorted(self._kinds.items())
+                ],
+                "objects": [meta.deep_copy(o) for b in self._store.values()
+                            for o in b.values()],
+            }
+
+    @classmethod
+    def load(cls, snapshot: dict) -> "FakeCluster":
+        """Rebuild a cluster from :meth:`dump` output.  resourceVersions are
+        reassigned (monotonic) but relative object content is preserved."""
+        cluster = cls()
+        for k in snapshot.get("kinds", []):
+            cluster.register_kind(k["apiVersion"], k["kind"], k["plural"],
+                                  k["namespaced"])
+        for obj in snapshot.get("objects", []):
+            obj = meta.deep_copy(obj)
+            obj.get("metadata", {}).pop("resourceVersion", None)
+            cluster.create(obj)
+        return cluster
 
     # -- convenience for tests/benchmarks ------------------------------------
 
