@@ -248,3 +248,50 @@ def test_status_subresource(rest, server):
     assert live["status"]["conditions"][0]["reason"] == "Ready"
     # spec untouched by the status patch
     assert live["spec"]["requestorID"] == "op"
+
+
+def test_from_environment_kubernetes_master(server, monkeypatch):
+    monkeypatch.delenv("KUBECONFIG", raising=False)
+    monkeypatch.setenv("KUBERNETES_MASTER", server.url)
+    c = RestClient.from_environment()
+    try:
+        c.create({"apiVersion": "v1", "kind": "Node",
+                  "metadata": {"name": "envn"}, "spec": {}})
+        assert c.get_node("envn")["metadata"]["name"] == "envn"
+    finally:
+        c.close()
+
+
+def test_from_environment_kubeconfig(server, tmp_path, monkeypatch):
+    kc = tmp_path / "kubeconfig"
+    kc.write_text(f"""
+apiVersion: v1
+kind: Config
+current-context: test
+contexts:
+  - name: test
+    context: {{cluster: c1, user: u1}}
+clusters:
+  - name: c1
+    cluster: {{server: "{server.url}", insecure-skip-tls-verify: true}}
+users:
+  - name: u1
+    user: {{token: dummy-token}}
+""")
+    monkeypatch.setenv("KUBECONFIG", str(kc))
+    monkeypatch.delenv("KUBERNETES_MASTER", raising=False)
+    c = RestClient.from_environment()
+    try:
+        assert c._http.headers["Authorization"] == "Bearer dummy-token"
+        c.create({"apiVersion": "v1", "kind": "Node",
+                  "metadata": {"name": "kcn"}, "spec": {}})
+        assert c.get_node("kcn")["metadata"]["name"] == "kcn"
+    finally:
+        c.close()
+
+
+def test_from_environment_nothing_configured(monkeypatch):
+    monkeypatch.delenv("KUBECONFIG", raising=False)
+    monkeypatch.delenv("KUBERNETES_MASTER", raising=False)
+    with pytest.raises(RuntimeError):
+        RestClient.from_environment()
