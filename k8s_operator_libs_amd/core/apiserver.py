@@ -27,6 +27,7 @@ def _status_body(exc: ApiError) -> dict:
         404: "NotFound",
         409: "Conflict",
         400: "BadRequest",
+        429: "TooManyRequests",
     }.get(exc.code, "InternalError")
     if exc.code == 409 and "already exists" in exc.message:
         reason = "AlreadyExists"

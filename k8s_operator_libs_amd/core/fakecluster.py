@@ -25,6 +25,7 @@ from typing import Callable, Dict, List, Optional, Tuple
 from . import meta
 from .errors import (
     AlreadyExistsError,
+    ApiError,
     BadRequestError,
     ConflictError,
     NotFoundError,
@@ -47,6 +48,7 @@ _BUILTIN_KINDS: Dict[Tuple[str, str], Tuple[str, bool]] = {
     # is maintenance.nvidia.com/v1alpha1 NodeMaintenance).
     ("maintenance.amd.com/v1alpha1", "NodeMaintenance"): ("nodemaintenances", True),
     ("coordination.k8s.io/v1", "Lease"): ("leases", True),
+    ("policy/v1", "PodDisruptionBudget"): ("poddisruptionbudgets", True),
 }
 
 
@@ -365,9 +367,63 @@ class FakeCluster:
     # -- pods ----------------------------------------------------------------
 
     def evict_pod(self, name: str, namespace: str) -> None:
-        """Eviction API: like envtest there is no kubelet, so eviction is an
-        immediate graceful delete."""
-        self.delete("v1", "Pod", name, namespace)
+        """Eviction API: enforces PodDisruptionBudgets like a real apiserver
+        (429 when the eviction would violate a budget), then deletes
+        immediately (no kubelet, like envtest)."""
+        with self._lock:
+            pod = self._bucket("v1", "Pod").get((namespace, name))
+            if pod is None:
+                raise NotFoundError(f"Pod {namespace}/{name} not found")
+            self._check_disruption_budgets(pod, namespace)
+            self.delete("v1", "Pod", name, namespace)
+
+    @staticmethod
+    def _pod_healthy(pod: K8sObject) -> bool:
+        if pod.get("status", {}).get("phase") != "Running":
+            return False
+        statuses = pod.get("status", {}).get("containerStatuses") or []
+        return bool(statuses) and all(c.get("ready") for c in statuses)
+
+    def _check_disruption_budgets(self, pod: K8sObject, namespace: str) -> None:
+        from .meta import match_labels_selector
+
+        pod_labels = pod.get("metadata", {}).get("labels", {}) or {}
+        for (ns, _), pdb in self._bucket("policy/v1", "PodDisruptionBudget").items():
+            if ns != namespace:
+                continue
+            sel = match_labels_selector(
+                pdb.get("spec", {}).get("selector", {}).get("matchLabels", {})
+            )
+            if not sel.matches(pod_labels):
+                continue
+            matched = [
+                p for (pns, _), p in self._bucket("v1", "Pod").items()
+                if pns == namespace and sel.matches(
+                    p.get("metadata", {}).get("labels", {}) or {})
+            ]
+            healthy = sum(1 for p in matched if self._pod_healthy(p))
+            spec = pdb.get("spec", {})
+            # Real PDBs budget against the owning controller's scale
+            # (status.expectedPods, maintained by the disruption controller);
+            # honour it when set, else fall back to the live matched count.
+            expected = pdb.get("status", {}).get("expectedPods") or len(matched)
+            min_available = spec.get("minAvailable")
+            max_unavailable = spec.get("maxUnavailable")
+            after = healthy - (1 if self._pod_healthy(pod) else 0)
+            blocked = False
+            if min_available is not None:
+                need = _scaled(min_available, expected, round_up=False)
+                blocked = after < need
+            elif max_unavailable is not None:
+                allowed = _scaled(max_unavailable, expected, round_up=False)
+                blocked = (expected - after) > allowed
+            if blocked:
+                err = ApiError(
+                    f"Cannot evict pod as it would violate the pod's "
+                    f"disruption budget {meta.name(pdb)}"
+                )
+                err.code = 429
+                raise err
 
     # -- watches -------------------------------------------------------------
 
@@ -422,6 +478,12 @@ class FakeCluster:
     def object_count(self) -> int:
         with self._lock:
             return sum(len(b) for b in self._store.values())
+
+
+def _scaled(value, total: int, round_up: bool) -> int:
+    from ..api.upgrade.v1alpha1 import IntOrString
+
+    return IntOrString.scaled_value(value, total, round_up)
 
 
 def _now_iso() -> str:

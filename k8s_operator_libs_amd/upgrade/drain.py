@@ -31,7 +31,7 @@ from typing import Callable, List, Optional
 from ..api.upgrade.v1alpha1 import DrainSpec
 from ..core import meta
 from ..core.client import Client
-from ..core.errors import NotFoundError
+from ..core.errors import ApiError, NotFoundError
 from . import consts
 
 logger = logging.getLogger(__name__)
@@ -169,17 +169,30 @@ def delete_or_evict_pods(
     to ``timeout_seconds``.
     """
     deadline = time.monotonic() + max(timeout_seconds, 0.001)
-    for pod in pods:
+
+    def try_evict(pod) -> bool:
+        """True when the evict/delete was accepted; False when a
+        PodDisruptionBudget blocked it (429) — retried until the deadline,
+        like kubectl drain."""
         try:
             if use_eviction:
                 client.evict_pod(meta.name(pod), meta.namespace(pod))
             else:
                 client.delete_pod(meta.name(pod), meta.namespace(pod))
+            return True
         except NotFoundError:
-            continue
+            return True
+        except ApiError as exc:
+            if exc.code == 429:
+                return False
+            raise
+
+    blocked = [pod for pod in pods if not try_evict(pod)]
     interval = 0.001
     remaining = list(pods)
     while remaining:
+        # PDB-blocked pods: another replica may have become healthy; retry
+        blocked = [pod for pod in blocked if not try_evict(pod)]
         still_there = []
         for pod in remaining:
             try:
@@ -194,6 +207,10 @@ def delete_or_evict_pods(
             return
         if time.monotonic() >= deadline:
             names = [f"{meta.namespace(p)}/{meta.name(p)}" for p in remaining]
+            if blocked:
+                names = [f"{n} (PDB-blocked)" if any(
+                    meta.name(b) == n.split("/")[-1] for b in blocked) else n
+                    for n in names]
             raise DrainError(f"timed out waiting for pods to terminate: {names}")
         time.sleep(interval)
         interval = min(interval * 2, 0.5)
