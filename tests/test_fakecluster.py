@@ -296,3 +296,43 @@ class TestDumpLoad:
             if state_of(clone_client, "node-0") == "upgrade-done":
                 break
         assert state_of(clone_client, "node-0") == "upgrade-done"
+
+
+class TestOptimisticConcurrencyUnderContention:
+    def test_rv_locked_patches_never_lose_updates(self, cluster):
+        """4 writers x 50 increments through RV-locked merge patches with
+        conflict retry: every increment must land (no lost updates)."""
+        import threading
+
+        cluster.create(mk_node("n1"))
+        cluster.patch("v1", "Node", "n1",
+                      {"metadata": {"annotations": {"counter": "0"}}})
+        errors = []
+
+        def writer():
+            try:
+                for _ in range(50):
+                    while True:
+                        live = cluster.get("v1", "Node", "n1")
+                        val = int(live["metadata"]["annotations"]["counter"])
+                        try:
+                            cluster.patch("v1", "Node", "n1", {
+                                "metadata": {
+                                    "resourceVersion": live["metadata"]["resourceVersion"],
+                                    "annotations": {"counter": str(val + 1)},
+                                }
+                            })
+                            break
+                        except ConflictError:
+                            continue
+            except Exception as exc:
+                errors.append(exc)
+
+        threads = [threading.Thread(target=writer) for _ in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+        assert not errors, errors
+        final = cluster.get("v1", "Node", "n1")["metadata"]["annotations"]["counter"]
+        assert final == "200"
