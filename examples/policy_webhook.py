@@ -12,8 +12,6 @@ reconcile error.
 POST /validate with an AdmissionReview v1; responds allowed=true/false.
 """
 
-from __future__ import annotations
-
 import argparse
 import sys
 
