@@ -15,7 +15,7 @@ import time
 
 sys.path.insert(0, __file__.rsplit("/", 2)[0])
 
-from k8s_operator_libs_amd.upgrade import consts, util
+from k8s_operator_libs_amd.upgrade import util
 
 
 def render(client, driver_name: str) -> str:

@@ -22,7 +22,7 @@ from __future__ import annotations
 
 import logging
 import threading
-from typing import Dict, Optional
+from typing import Dict
 
 from ..api.upgrade.v1alpha1 import DriverUpgradePolicySpec
 from ..core import meta
