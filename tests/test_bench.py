@@ -45,3 +45,16 @@ def test_bench_requestor_mode():
         print_json=False, mode="requestor",
     )
     assert result["upgrades_completed"] == 1
+
+
+def test_performance_regression_guard():
+    """Generous ceiling to catch pathological slowdowns in CI: a full 8-node
+    rolling upgrade (live reconcile, no GPU) must finish well under 2s
+    (measured ~5-25ms on healthy machines)."""
+    import bench
+
+    result = bench.run_rolling_upgrade_benchmark(
+        n_nodes=8, steps=3, warmup=1, gpu_validate=False, print_json=False,
+    )
+    assert result["mean_wall_s"] < 2.0, result
+    assert result["reconcile_p50_ms"] < 200.0, result
