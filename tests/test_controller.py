@@ -63,3 +63,39 @@ def test_status_cli_renders(client):
     assert "node-0" in out and "upgrade-done" in out
     assert "drain-required" in out and "cordoned" in out
     assert "totals:" in out
+
+
+def test_requestor_controller_wakes_on_nodemaintenance_ready(client):
+    """In requestor mode the controller subscribes to NodeMaintenance
+    condition changes (filtered by requestor-ID + condition predicates) and
+    completes the upgrade without relying on the resync timer."""
+    from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
+    from k8s_operator_libs_amd.upgrade.state_manager import StateOptions
+    from simenv import SimMaintenanceOperator
+
+    ds, _ = setup_cluster(client, pod_hash="old", ds_hash="new")
+    SimDaemonSetController(client.cluster, ds, current_hash="new")
+    SimMaintenanceOperator(client.cluster)
+    manager = ClusterUpgradeStateManager(
+        client,
+        options=StateOptions(requestor=RequestorOptions(
+            use_maintenance_operator=True,
+            requestor_id="amd.gpu.operator",
+            namespace="default",
+        )),
+    )
+    controller = UpgradeController(
+        manager, DRIVER_NS, DRIVER_LABELS, policy(drainSpec={"enable": True}),
+        resync_seconds=120.0,  # far beyond the test window: events must drive it
+    )
+    done = {}
+    t = threading.Thread(
+        target=lambda: done.update(ok=controller.run(until_all_done=True,
+                                                     max_reconciles=30)),
+        daemon=True,
+    )
+    t.start()
+    t.join(timeout=25)
+    controller.stop()
+    assert done.get("ok") is True
+    assert state_of(client, "node-0") == consts.UPGRADE_STATE_DONE
