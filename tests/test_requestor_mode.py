@@ -316,3 +316,73 @@ class TestPredicates:
         new = {"metadata": {"finalizers": [], "deletionTimestamp": "t"},
                "status": {"conditions": []}}
         assert condition_changed_predicate(old, new)
+
+
+class TestSharedWindowEndToEnd:
+    def test_gpu_and_nic_operators_share_one_maintenance_window(self, client):
+        """The reference's marquee multi-operator flow
+        (docs/automatic-ofed-upgrade.md:117-135): GPU and NIC operators with
+        the default NodeMaintenance prefix share ONE maintenance window per
+        node — the second operator joins additionalRequestors instead of
+        creating a duplicate, each removes itself on completion, and the
+        owner's deletion releases the node."""
+        from simenv import SimMaintenanceOperator
+
+        nm_name = f"{DEFAULT_NODE_MAINTENANCE_NAME_PREFIX}-node-0"
+        SimMaintenanceOperator(client.cluster)
+
+        # amdgpu driver DS (out of date) under driver name "amdgpu"
+        util.set_driver_name("amdgpu")
+        gpu_ds, _ = setup_cluster(client, pod_hash="old", ds_hash="new",
+                                  ds_name="amdgpu-driver")
+        SimDaemonSetController(client.cluster, gpu_ds, current_hash="new")
+        gpu_mgr = make_manager(client, requestor_id="amd.gpu.operator")
+
+        # anic driver DS (out of date) on the SAME node, driver name "anic"
+        util.set_driver_name("anic")
+        from builders import DaemonSetBuilder, driver_pod_for, make_controller_revision
+
+        nic_labels = {"app": "anic-driver-daemonset"}
+        nic_ds = DaemonSetBuilder("anic-driver", labels=nic_labels) \
+            .with_desired_number_scheduled(1).build(client.cluster)
+        make_controller_revision(nic_ds, "new", revision=2, cluster=client.cluster)
+        make_controller_revision(nic_ds, "old", revision=1, cluster=client.cluster)
+        driver_pod_for(nic_ds, "node-0", hash_="old").build(client.cluster)
+        SimDaemonSetController(client.cluster, nic_ds, current_hash="new")
+        nic_mgr = make_manager(client, requestor_id="amd.network.operator")
+
+        pol = policy(drainSpec={"enable": True})
+        saw_shared = False
+        for _ in range(15):
+            util.set_driver_name("amdgpu")
+            gpu_mgr.reconcile(DRIVER_NS, DRIVER_LABELS, pol)
+            util.set_driver_name("anic")
+            nic_mgr.reconcile(DRIVER_NS, nic_labels, pol)
+            try:
+                nm = client.get(NM_API, "NodeMaintenance", nm_name, "default")
+                if nm["spec"].get("additionalRequestors"):
+                    saw_shared = True
+                    # one object, two requestors — never two objects
+                    assert nm["spec"]["requestorID"] in (
+                        "amd.gpu.operator", "amd.network.operator")
+            except NotFoundError:
+                pass
+            util.set_driver_name("amdgpu")
+            gpu_done = state_of(client, "node-0") == consts.UPGRADE_STATE_DONE
+            util.set_driver_name("anic")
+            nic_done = state_of(client, "node-0") == consts.UPGRADE_STATE_DONE
+            if gpu_done and nic_done:
+                break
+        assert gpu_done and nic_done, (gpu_done, nic_done)
+        assert saw_shared, "operators never shared the maintenance window"
+        # window fully released: object gone, node schedulable
+        with pytest.raises(NotFoundError):
+            client.get(NM_API, "NodeMaintenance", nm_name, "default")
+        assert not client.get_node("node-0")["spec"].get("unschedulable")
+        # both drivers on their new revisions
+        for sel, ns_ in (("app=amdgpu-driver-daemonset", DRIVER_NS),
+                         ("app=anic-driver-daemonset", DRIVER_NS)):
+            pods = client.list_pods(namespace=ns_, label_selector=sel)
+            assert pods and all(
+                p["metadata"]["labels"]["controller-revision-hash"] == "new"
+                for p in pods)
