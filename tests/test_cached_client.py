@@ -85,7 +85,9 @@ def test_full_upgrade_over_lagging_cache(client):
         manager = ClusterUpgradeStateManager(cached)
         pol = policy(maxParallelUpgrades=1, maxUnavailable="100%")
         transitions = manager.metrics.state_transitions
-        for _ in range(12):
+        # generous budget: under parallel test load the watch thread gets
+        # scheduled late and each tick may deliver nothing yet
+        for _ in range(30):
             state = manager.build_state(DRIVER_NS, DRIVER_LABELS)
             manager.apply_state(state, pol)
             manager.wait_idle()
