@@ -164,3 +164,13 @@ def test_log_level_mapping():
     assert consts.to_logging_level(consts.LOG_LEVEL_INFO) == logging.INFO
     assert consts.to_logging_level(5) == logging.DEBUG
     assert consts.to_logging_level(-9) == logging.ERROR
+
+
+def test_testing_module_surface():
+    from k8s_operator_libs_amd import testing
+
+    for name in ("NodeBuilder", "PodBuilder", "DaemonSetBuilder",
+                 "NodeMaintenanceBuilder", "SimDaemonSetController",
+                 "SimMaintenanceOperator", "driver_pod_for",
+                 "make_controller_revision"):
+        assert hasattr(testing, name), name
