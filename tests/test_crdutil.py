@@ -188,3 +188,77 @@ def test_generated_consumer_crd_applies(client, tmp_path):
     })
     got = client.get("driver.amd.com/v1alpha1", "AMDGPUDriver", "default")
     assert got["spec"]["driverUpgradePolicy"]["maxParallelUpgrades"] == 2
+
+
+class TestApplyRaces:
+    def test_update_conflict_retries_until_success(self, client, tmp_path):
+        """A concurrent writer bumping the CRD between our GET and UPDATE
+        forces RetryOnConflict behaviour (crdutil.go:214-249)."""
+        from k8s_operator_libs_amd.crdutil import crdutil as crdmod
+
+        p = write(str(tmp_path / "w.yaml"),
+                  CRD_TMPL.format(group="amd.com", kind="Widget", plural="widgets"))
+        process_crds(client, [p], CRD_OPERATION_APPLY)
+
+        real_update = client.update
+        bumps = {"n": 0}
+
+        def racing_update(obj):
+            if bumps["n"] < 2:
+                bumps["n"] += 1
+                # concurrent writer invalidates our resourceVersion
+                live = client.get("apiextensions.k8s.io/v1",
+                                  "CustomResourceDefinition", "widgets.amd.com")
+                live["metadata"]["labels"] = {"race": str(bumps["n"])}
+                real_update(live)
+            return real_update(obj)
+
+        client.update = racing_update
+        try:
+            updated = CRD_TMPL.format(group="amd.com", kind="Widget",
+                                      plural="widgets").replace(
+                "metadata:\n  name: widgets.amd.com",
+                "metadata:\n  name: widgets.amd.com\n  labels:\n    final: 'yes'")
+            p2 = write(str(tmp_path / "w2.yaml"), updated)
+            process_crds(client, [p2], CRD_OPERATION_APPLY)
+        finally:
+            client.update = real_update
+        got = client.get("apiextensions.k8s.io/v1", "CustomResourceDefinition",
+                         "widgets.amd.com")
+        assert got["metadata"]["labels"]["final"] == "yes"
+        assert bumps["n"] == 2  # retried through both injected conflicts
+
+    def test_create_race_falls_back_to_update(self, client, tmp_path):
+        """Another applier creating the CRD between our GET (404) and CREATE
+        must turn into an update, not an error (crdutil.go:224-231)."""
+        p = write(str(tmp_path / "w.yaml"),
+                  CRD_TMPL.format(group="amd.com", kind="Widget", plural="widgets"))
+        real_create = client.create
+        raced = {"done": False}
+
+        def racing_create(obj):
+            if obj.get("kind") == "CustomResourceDefinition" and not raced["done"]:
+                raced["done"] = True
+                real_create(obj)  # the other applier wins the create
+            return real_create(obj)  # ours now raises AlreadyExists
+
+        client.create = racing_create
+        try:
+            process_crds(client, [p], CRD_OPERATION_APPLY)
+        finally:
+            client.create = real_create
+        assert client.get("apiextensions.k8s.io/v1", "CustomResourceDefinition",
+                          "widgets.amd.com")
+
+    def test_wait_for_crds_times_out_on_never_served(self, client):
+        from k8s_operator_libs_amd.crdutil import CrdUtilError, wait_for_crds
+
+        ghost = {"apiVersion": "apiextensions.k8s.io/v1",
+                 "kind": "CustomResourceDefinition",
+                 "metadata": {"name": "ghosts.amd.com"},
+                 "spec": {"group": "amd.com", "scope": "Namespaced",
+                          "names": {"kind": "Ghost", "plural": "ghosts"},
+                          "versions": [{"name": "v1", "served": True}]}}
+        # never created on the cluster: discovery can't serve it
+        with pytest.raises(CrdUtilError):
+            wait_for_crds(client, [ghost], timeout=0.3)
