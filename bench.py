@@ -232,7 +232,7 @@ def _run_gpu_validation(device):
 
 def run_rolling_upgrade_benchmark(
     n_nodes=8, steps=10, warmup=2, max_parallel=2, gpu_validate=False,
-    device=0, gpu_pods_per_node=2, print_json=True, converge=True,
+    device=0, gpu_pods_per_node=2, converge=True,
     mode="inplace",
 ):
     """Run `warmup` untimed + `steps` timed full rolling upgrades; returns a
@@ -376,7 +376,7 @@ def main():
     run_rolling_upgrade_benchmark(
         n_nodes=args.nodes, steps=0, warmup=args.warmup,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
-        device=local_rank if have_cuda else 0, print_json=False, converge=converge,
+        device=local_rank if have_cuda else 0, converge=converge,
         mode=args.mode,
     )
     barrier_sync()
@@ -384,7 +384,7 @@ def main():
     result = run_rolling_upgrade_benchmark(
         n_nodes=args.nodes, steps=args.steps, warmup=0,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
-        device=local_rank if have_cuda else 0, print_json=False, converge=converge,
+        device=local_rank if have_cuda else 0, converge=converge,
         mode=args.mode,
     )
     barrier_sync()

@@ -42,7 +42,7 @@ def test_bench_requestor_mode():
 
     result = bench.run_rolling_upgrade_benchmark(
         n_nodes=2, steps=1, warmup=0, max_parallel=1, gpu_validate=False,
-        print_json=False, mode="requestor",
+        mode="requestor",
     )
     assert result["upgrades_completed"] == 1
 
@@ -54,7 +54,7 @@ def test_performance_regression_guard():
     import bench
 
     result = bench.run_rolling_upgrade_benchmark(
-        n_nodes=8, steps=3, warmup=1, gpu_validate=False, print_json=False,
+        n_nodes=8, steps=3, warmup=1, gpu_validate=False,
     )
     assert result["mean_wall_s"] < 2.0, result
     assert result["reconcile_p50_ms"] < 200.0, result
