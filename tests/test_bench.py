@@ -12,7 +12,6 @@ def test_rolling_upgrade_benchmark_function():
 
     result = bench.run_rolling_upgrade_benchmark(
         n_nodes=2, steps=1, warmup=0, max_parallel=1, gpu_validate=False,
-        print_json=False,
     )
     assert result["upgrades_completed"] == 1
     assert result["mean_wall_s"] > 0
