@@ -60,3 +60,48 @@ def test_events_recorded(client):
     node = NodeBuilder("n1").build(client.cluster)
     provider.change_node_upgrade_state(node, consts.UPGRADE_STATE_UPGRADE_REQUIRED)
     assert any("Successfully updated node state label" in e for e in rec.events)
+
+
+def test_patch_failure_records_warning_event_and_raises(client):
+    class FailingClient:
+        cluster = client.cluster
+
+        def __getattr__(self, name):
+            return getattr(client, name)
+
+        def patch(self, *a, **kw):
+            raise RuntimeError("apiserver unavailable")
+
+    rec = FakeRecorder()
+    provider = NodeUpgradeStateProvider(FailingClient(), rec)
+    node = NodeBuilder("n1").build(client.cluster)
+    with pytest.raises(RuntimeError):
+        provider.change_node_upgrade_state(node, consts.UPGRADE_STATE_UPGRADE_REQUIRED)
+    assert any("Failed to update node state label" in e for e in rec.events)
+    # the in-memory snapshot was NOT mutated on failure
+    assert util.get_upgrade_state_label_key() not in node["metadata"]["labels"]
+
+
+def test_barrier_timeout_when_cache_never_converges(client, monkeypatch):
+    """A cache that never reflects the patch must raise after the deadline
+    rather than return (double-fire protection)."""
+    from k8s_operator_libs_amd.upgrade import node_state_provider as nsp
+
+    class FrozenCacheClient:
+        cluster = client.cluster
+
+        def __getattr__(self, name):
+            return getattr(client, name)
+
+        def patch(self, *a, **kw):
+            return client.patch(*a, **kw)
+
+        def get_node(self, name):
+            # always serve the ORIGINAL (stale) object
+            return {"metadata": {"name": name, "labels": {}, "annotations": {}}}
+
+    monkeypatch.setattr(nsp, "_BARRIER_TIMEOUT_S", 0.2)
+    provider = NodeUpgradeStateProvider(FrozenCacheClient())
+    node = NodeBuilder("n1").build(client.cluster)
+    with pytest.raises(nsp.StateChangeTimeoutError):
+        provider.change_node_upgrade_state(node, consts.UPGRADE_STATE_UPGRADE_REQUIRED)
