@@ -159,6 +159,53 @@ static uint8_t encode_e4m3fn_nearest(float f) {
   return best;
 }
 
+static py::dict mfma_bf16_tile(int device) {
+  // Run the bf16 MFMA tile and return decoded inputs + GPU output so Python
+  // tests can verify against an independent fp32 reference (e.g. torch).
+  HIP_CHECK(hipSetDevice(device));
+  const int M = 16, N = 16, K = 32;
+  std::vector<float> hAf(M * K), hBf(K * N), hD(M * N);
+  std::vector<uint16_t> hA(M * K), hB(K * N);
+  auto to_bf16 = [](float f) -> uint16_t {
+    uint32_t u;
+    std::memcpy(&u, &f, 4);
+    uint32_t rounded = u + 0x7FFF + ((u >> 16) & 1);
+    return (uint16_t)(rounded >> 16);
+  };
+  auto from_bf16 = [](uint16_t v) -> float {
+    uint32_t u = (uint32_t)v << 16;
+    float f;
+    std::memcpy(&f, &u, 4);
+    return f;
+  };
+  for (int i = 0; i < M * K; ++i) hA[i] = to_bf16(0.07f * (float)((i * 13) % 41) - 1.2f);
+  for (int i = 0; i < K * N; ++i) hB[i] = to_bf16(0.05f * (float)((i * 17) % 37) - 0.8f);
+  for (int i = 0; i < M * K; ++i) hAf[i] = from_bf16(hA[i]);
+  for (int i = 0; i < K * N; ++i) hBf[i] = from_bf16(hB[i]);
+  uint16_t *dA, *dB;
+  float* dD;
+  HIP_CHECK(hipMalloc(&dA, sizeof(uint16_t) * M * K));
+  HIP_CHECK(hipMalloc(&dB, sizeof(uint16_t) * K * N));
+  HIP_CHECK(hipMalloc(&dD, sizeof(float) * M * N));
+  HIP_CHECK(hipMemcpy(dA, hA.data(), sizeof(uint16_t) * M * K, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(dB, hB.data(), sizeof(uint16_t) * K * N, hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(mfma_bf16_16x16x32_kernel, dim3(1), dim3(64), 0, 0,
+                     (const __bf16*)dA, (const __bf16*)dB, dD);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipMemcpy(hD.data(), dD, sizeof(float) * M * N, hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(dA));
+  HIP_CHECK(hipFree(dB));
+  HIP_CHECK(hipFree(dD));
+  py::dict out;
+  out["m"] = M;
+  out["n"] = N;
+  out["k"] = K;
+  out["a"] = hAf;  // bf16-quantized values as f32, row-major MxK
+  out["b"] = hBf;  // row-major KxN
+  out["d"] = hD;   // GPU MFMA result, row-major MxN
+  return out;
+}
+
 static double mfma_fp8_check(int device) {
   HIP_CHECK(hipSetDevice(device));
   const int M = 16, N = 16, K = 32;
@@ -540,6 +587,9 @@ PYBIND11_MODULE(_gpu_validator, m) {
         py::arg("buf_mib") = 1024.0, py::arg("iters") = 10,
         "Streaming float4 copy bandwidth in GB/s (read+write)");
   m.def("lds_roundtrip_check", &lds_roundtrip_check, py::arg("device") = 0);
+  m.def("mfma_bf16_tile", &mfma_bf16_tile, py::arg("device") = 0,
+        "Run one bf16 MFMA tile; returns quantized inputs + GPU output for "
+        "independent (e.g. torch fp32) verification");
   m.def("mfma_fp8_check", &mfma_fp8_check, py::arg("device") = 0,
         "Max abs error of a v_mfma_f32_16x16x32_fp8_fp8 (OCP e4m3) tile vs "
         "f32 CPU reference");

@@ -94,3 +94,19 @@ def test_deep_health_check_on_gpu():
     assert report["healthy"], report
     assert report["checks"]["mfma_fp8_max_err"] <= 5e-2
     assert report["checks"]["mfma_throughput_tflops"] > 1000.0
+
+
+def test_mfma_bf16_vs_torch_fp32_reference(native):
+    """Numerics contract: the HIP MFMA kernel vs a plain PyTorch fp32
+    reference of the same op (matmul over the bf16-quantized operands)."""
+    import torch
+
+    tile = native.mfma_bf16_tile(0)
+    m, n, k = tile["m"], tile["n"], tile["k"]
+    a = torch.tensor(tile["a"], dtype=torch.float32).reshape(m, k)
+    b = torch.tensor(tile["b"], dtype=torch.float32).reshape(k, n)
+    d_gpu = torch.tensor(tile["d"], dtype=torch.float32).reshape(m, n)
+    d_ref = a @ b
+    assert torch.allclose(d_gpu, d_ref, rtol=1e-5, atol=1e-4), (
+        (d_gpu - d_ref).abs().max().item()
+    )
