@@ -295,3 +295,37 @@ def test_from_environment_nothing_configured(monkeypatch):
     monkeypatch.delenv("KUBERNETES_MASTER", raising=False)
     with pytest.raises(RuntimeError):
         RestClient.from_environment()
+
+
+def test_full_upgrade_through_informer_stack(rest, server):
+    """The complete production architecture in one test: state machine ->
+    CachedClient (watch-fed informers) -> RestClient -> HTTP -> apiserver.
+    Reads come from the cache, writes go through REST, the provider barrier
+    holds transitions until the informers converge."""
+    import time
+
+    from k8s_operator_libs_amd.core.cache import CachedClient
+
+    class W:
+        cluster = server.cluster
+
+    ds, _ = setup_cluster(W, pod_hash="old", ds_hash="new")
+    SimDaemonSetController(server.cluster, ds, current_hash="new")
+    cached = CachedClient(rest)
+    try:
+        manager = ClusterUpgradeStateManager(cached).with_pod_deletion_enabled(
+            gpu_pod_deletion_filter
+        )
+        pol = policy(maxParallelUpgrades=1, maxUnavailable="100%",
+                     drainSpec={"enable": True})
+        for _ in range(25):
+            manager.reconcile(DRIVER_NS, DRIVER_LABELS, pol)
+            if state_of(rest, "node-0") == consts.UPGRADE_STATE_DONE:
+                break
+            time.sleep(0.1)  # informer requeue interval
+        assert state_of(rest, "node-0") == consts.UPGRADE_STATE_DONE
+        # every label transition fired exactly once despite the async cache
+        for (frm, to), count in manager.metrics.state_transitions.items().items():
+            assert count == 1, f"{frm}->{to} fired {count} times"
+    finally:
+        cached.stop()
