@@ -80,8 +80,20 @@ def run_demo(args):
     ds = bench_mod._make_cluster(W, args.demo_nodes, "oldrev", "newrev")
     bench_mod._DsController(handle.cluster, ds, "newrev")
 
+    options = None
+    if args.demo_requestor:
+        from k8s_operator_libs_amd.testing import SimMaintenanceOperator
+        from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
+
+        SimMaintenanceOperator(handle.cluster)
+        options = StateOptions(requestor=RequestorOptions(
+            use_maintenance_operator=True,
+            requestor_id="amd.gpu.operator",
+            namespace="default",
+        ))
+        log.info("demo: requestor mode (simulated maintenance operator)")
     manager = (
-        ClusterUpgradeStateManager(client)
+        ClusterUpgradeStateManager(client, options=options)
         .with_pod_deletion_enabled(gpu_pod_deletion_filter)
         .with_validation_enabled("app=amd-gpu-validator")
     )
@@ -145,6 +157,8 @@ def main(argv=None) -> int:
     parser.add_argument("--demo", action="store_true")
     parser.add_argument("--demo-nodes", type=int, default=8)
     parser.add_argument("--no-leader-election", action="store_true")
+    parser.add_argument("--demo-requestor", action="store_true",
+                        help="demo: delegate node quiescing to a simulated maintenance operator")
     args = parser.parse_args(argv)
 
     logging.basicConfig(level=logging.INFO,

@@ -82,3 +82,11 @@ class TestPolicyWebhook:
         finally:
             server.should_exit = True
             t.join(timeout=5)
+
+
+def test_demo_requestor_mode_completes():
+    rc = operator.main([
+        "--demo", "--demo-requestor", "--demo-nodes", "2",
+        "--interval", "0.02", "--metrics-port", "18879",
+    ])
+    assert rc == 0
