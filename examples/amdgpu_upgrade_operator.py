@@ -85,7 +85,13 @@ def run_demo(args):
         from k8s_operator_libs_amd.testing import SimMaintenanceOperator
         from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
 
-        SimMaintenanceOperator(handle.cluster)
+        # the simulated maintenance operator must not evict the validator
+        # pods the upgrade gates on (the real one applies podEvictionFilters)
+        SimMaintenanceOperator(
+            handle.cluster,
+            evict_filter=lambda pod: pod["metadata"].get("labels", {})
+            .get("app") != "amd-gpu-validator",
+        )
         options = StateOptions(requestor=RequestorOptions(
             use_maintenance_operator=True,
             requestor_id="amd.gpu.operator",

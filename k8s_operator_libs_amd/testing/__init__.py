@@ -255,9 +255,13 @@ class SimMaintenanceOperator:
 
     FINALIZER = "maintenance.amd.com/guard"
 
-    def __init__(self, cluster, evict=True):
+    def __init__(self, cluster, evict=True, evict_filter=None):
+        """``evict_filter(pod) -> bool``: which non-DaemonSet pods to evict
+        during maintenance (the real operator applies the NodeMaintenance
+        spec's podEvictionFilters); default evicts all of them."""
         self.cluster = cluster
         self.evict = evict
+        self.evict_filter = evict_filter
         self._lock = threading.RLock()
         cluster.add_change_hook(self._on_change)
 
@@ -309,6 +313,8 @@ class SimMaintenanceOperator:
                             ):
                                 refs = meta.owner_references(pod)
                                 if refs and refs[0].get("kind") == "DaemonSet":
+                                    continue
+                                if self.evict_filter is not None and                                         not self.evict_filter(pod):
                                     continue
                                 try:
                                     self.cluster.delete("v1", "Pod", meta.name(pod),
