@@ -286,13 +286,13 @@ class _HttpWatch:
         self._stop = threading.Event()
         self._connected = threading.Event()
         path = client._collection_path(api_version, kind, "")
-        url = client.base_url + path
 
         def reader():
             try:
-                with httpx.stream(
-                    "GET", url, params={"watch": "true"},
-                    headers=dict(client._http.headers), timeout=None,
+                # stream through the client's own session so TLS verification,
+                # auth headers and base_url apply to watches too
+                with client._http.stream(
+                    "GET", path, params={"watch": "true"}, timeout=None,
                 ) as resp:
                     for line in resp.iter_lines():
                         if self._stop.is_set():
