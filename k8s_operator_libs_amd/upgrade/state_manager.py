@@ -99,8 +99,11 @@ class ClusterUpgradeStateManager:
         t0 = time.perf_counter()
         state = self._build_state(namespace, driver_labels)
         self.metrics.build_state_duration.observe(time.perf_counter() - t0)
-        for state_name, node_states in state.node_states.items():
-            self.metrics.node_states.set(len(node_states), state_name)
+        # zero absent states so the gauge never reports a stale count
+        for state_name in consts.ALL_STATES:
+            self.metrics.node_states.set(
+                len(state.nodes_in(state_name)), state_name
+            )
         return state
 
     def _build_state(self, namespace: str, driver_labels: Dict[str, str]) -> ClusterUpgradeState:

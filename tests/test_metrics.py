@@ -85,3 +85,21 @@ def test_prometheus_text_exposition(client):
     families = list(text_string_to_metric_families(text))
     names = {f.name for f in families}
     assert "amd_upgrade_reconcile_duration_seconds" in names
+
+
+def test_node_state_gauge_zeroes_on_transition(client):
+    from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
+
+    reg = MetricsRegistry()
+    manager = ClusterUpgradeStateManager(client, metrics=reg)
+    setup_cluster(client, node_states=consts.UPGRADE_STATE_DRAIN_REQUIRED)
+    manager.build_state(DRIVER_NS, DRIVER_LABELS)
+    assert reg.node_states.value(consts.UPGRADE_STATE_DRAIN_REQUIRED) == 1
+    # node moves on; the old state's gauge must drop to 0, not linger
+    node = client.get_node("node-0")
+    manager.common.node_state_provider.change_node_upgrade_state(
+        node, consts.UPGRADE_STATE_DONE
+    )
+    manager.build_state(DRIVER_NS, DRIVER_LABELS)
+    assert reg.node_states.value(consts.UPGRADE_STATE_DRAIN_REQUIRED) == 0
+    assert reg.node_states.value(consts.UPGRADE_STATE_DONE) == 1
