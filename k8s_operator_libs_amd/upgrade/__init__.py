@@ -22,6 +22,7 @@ from .common_manager import (  # noqa: F401
     is_node_unschedulable,
     is_orphaned_pod,
 )
+from .controller import UpgradeController  # noqa: F401
 from .cordon_manager import CordonManager  # noqa: F401
 from .drain import gpu_pod_deletion_filter, pod_requests_resource  # noqa: F401
 from .drain_manager import DrainConfiguration, DrainManager  # noqa: F401

@@ -174,3 +174,9 @@ def test_testing_module_surface():
                  "SimMaintenanceOperator", "driver_pod_for",
                  "make_controller_revision"):
         assert hasattr(testing, name), name
+
+
+def test_controller_exported_from_upgrade_package():
+    from k8s_operator_libs_amd import upgrade
+
+    assert upgrade.UpgradeController
