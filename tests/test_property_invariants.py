@@ -94,3 +94,62 @@ def test_int_or_percent_bounds(total, pct):
     assert down <= up <= down + 1
     assert up == math.ceil(pct * total / 100)
     assert 0 <= up <= total or pct > 100
+
+
+_key = st.text(alphabet="abcz-._/", min_size=1, max_size=10).filter(
+    lambda s: s[0].isalnum() or s[0] in "abcz"
+)
+_val = st.text(alphabet="abcz123-", min_size=0, max_size=8)
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    reqs=st.lists(
+        st.one_of(
+            st.tuples(st.just("eq"), _key, _val),
+            st.tuples(st.just("neq"), _key, _val),
+            st.tuples(st.just("exists"), _key, st.just("")),
+            st.tuples(st.just("notexists"), _key, st.just("")),
+            st.tuples(st.just("in"), _key,
+                      st.lists(_val.filter(bool), min_size=1, max_size=3)),
+        ),
+        min_size=1, max_size=4,
+    ),
+    labels=st.dictionaries(_key, _val, max_size=4),
+)
+def test_label_selector_roundtrip_matches_composed_predicate(reqs, labels):
+    """Build a selector string from structured requirements, parse it, and
+    check matches() against the independently-composed predicate."""
+    from k8s_operator_libs_amd.core.meta import LabelSelector
+
+    parts, preds = [], []
+    for op, key, val in reqs:
+        if op == "eq":
+            parts.append(f"{key}={val}")
+            preds.append(lambda l, k=key, v=val: l.get(k) == v)
+        elif op == "neq":
+            parts.append(f"{key}!={val}")
+            preds.append(lambda l, k=key, v=val: not (k in l and l[k] == v))
+        elif op == "exists":
+            parts.append(key)
+            preds.append(lambda l, k=key: k in l)
+        elif op == "notexists":
+            parts.append(f"!{key}")
+            preds.append(lambda l, k=key: k not in l)
+        elif op == "in":
+            parts.append(f"{key} in ({','.join(val)})")
+            preds.append(lambda l, k=key, vs=set(val): l.get(k) in vs)
+    selector = ",".join(parts)
+    sel = LabelSelector(selector)
+    expected = all(p(labels) for p in preds)
+    assert sel.matches(labels) == expected, (selector, labels)
+
+
+@settings(max_examples=60, deadline=None)
+@given(junk=st.text(max_size=40))
+def test_label_selector_never_crashes_on_junk(junk):
+    from k8s_operator_libs_amd.core.meta import LabelSelector
+
+    sel = LabelSelector(junk)
+    sel.matches({"a": "b"})
+    sel.matches({})
