@@ -153,3 +153,43 @@ def test_label_selector_never_crashes_on_junk(junk):
     sel = LabelSelector(junk)
     sel.matches({"a": "b"})
     sel.matches({})
+
+
+_json_scalar = st.one_of(st.none(), st.booleans(), st.integers(-9, 9),
+                         st.text(alphabet="xyz", max_size=3))
+_json = st.recursive(
+    _json_scalar,
+    lambda children: st.one_of(
+        st.lists(children, max_size=3),
+        st.dictionaries(st.text(alphabet="abc", min_size=1, max_size=2),
+                        children, max_size=3),
+    ),
+    max_leaves=12,
+)
+
+
+def _spec_merge_patch(target, patch):
+    """RFC 7386 pseudocode, transcribed independently of the implementation."""
+    if isinstance(patch, dict):
+        if not isinstance(target, dict):
+            target = {}
+        result = dict(target)
+        for k, v in patch.items():
+            if v is None:
+                result.pop(k, None)
+            else:
+                result[k] = _spec_merge_patch(result.get(k), v)
+        return result
+    return patch
+
+
+@settings(max_examples=120, deadline=None)
+@given(target=_json, patch=_json)
+def test_merge_patch_matches_rfc_pseudocode(target, patch):
+    import copy as _copy
+
+    from k8s_operator_libs_amd.core.meta import json_merge_patch
+
+    expected = _spec_merge_patch(_copy.deepcopy(target), patch)
+    got = json_merge_patch(_copy.deepcopy(target), patch)
+    assert got == expected
