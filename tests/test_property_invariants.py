@@ -193,3 +193,31 @@ def test_merge_patch_matches_rfc_pseudocode(target, patch):
     expected = _spec_merge_patch(_copy.deepcopy(target), patch)
     got = json_merge_patch(_copy.deepcopy(target), patch)
     assert got == expected
+
+
+@settings(max_examples=120, deadline=None)
+@given(tree=_json)
+def test_native_deep_copy_matches_stdlib(tree):
+    """The C++ _jsonops.deep_copy must agree with copy.deepcopy on arbitrary
+    JSON trees and produce fully independent containers."""
+    import copy as _copy
+
+    from k8s_operator_libs_amd.core import meta
+
+    got = meta.deep_copy(tree)
+    assert got == _copy.deepcopy(tree)
+    # independence: mutate every dict/list in the copy; original unchanged
+    snapshot = _copy.deepcopy(tree)
+
+    def mutate(node):
+        if isinstance(node, dict):
+            node["__mut__"] = 1
+            for v in list(node.values()):
+                mutate(v)
+        elif isinstance(node, list):
+            node.append("__mut__")
+            for v in node[:-1]:
+                mutate(v)
+
+    mutate(got)
+    assert tree == snapshot
