@@ -18,6 +18,14 @@ directly (CPU-only mode, reported in the JSON).
 
 Scaling is weak: each rank runs an independent 8-node cluster upgrade
 (one rank per GPU over torch.distributed, rendezvous provided by the driver).
+
+Flags beyond the driver contract (--gpus/--steps/--warmup):
+  --nodes N        simulated cluster size per rank (default 8)
+  --max-parallel   rolling window (default 2, the BASELINE config)
+  --mode           inplace (default) | requestor (maintenance-operator
+                   delegation with a simulated maintenance operator)
+  --no-converge    strict reference semantics (one transition per tick)
+  --no-gpu-validate  skip the native MFMA check in the validation step
 """
 
 from __future__ import annotations
