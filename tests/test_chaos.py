@@ -10,6 +10,12 @@ import random
 
 from hypothesis import given, settings, strategies as st
 
+# Scale example counts for deep campaigns: HYPOTHESIS_SCALE=20 multiplies
+# every property's max_examples (default 1).
+import os as _os
+
+_SCALE = max(1, int(_os.environ.get("HYPOTHESIS_SCALE", "1")))
+
 from k8s_operator_libs_amd.core import FakeClient
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
@@ -47,7 +53,7 @@ def _inject_fault(rng, client, n_nodes):
         pass  # faults racing each other is part of the chaos
 
 
-@settings(max_examples=12, deadline=None)
+@settings(max_examples=12 * _SCALE, deadline=None)
 @given(seed=st.integers(min_value=0, max_value=10_000))
 def test_converges_despite_random_faults(seed):
     rng = random.Random(seed)

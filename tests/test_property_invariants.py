@@ -13,6 +13,12 @@ import math
 
 from hypothesis import given, settings, strategies as st
 
+# Scale example counts for deep campaigns: HYPOTHESIS_SCALE=20 multiplies
+# every property's max_examples (default 1).
+import os as _os
+
+_SCALE = max(1, int(_os.environ.get("HYPOTHESIS_SCALE", "1")))
+
 from k8s_operator_libs_amd.api.upgrade.v1alpha1 import IntOrString
 from k8s_operator_libs_amd.core import FakeClient
 from k8s_operator_libs_amd.metrics import MetricsRegistry
@@ -23,7 +29,7 @@ from simenv import SimDaemonSetController
 from test_state_manager import policy, setup_cluster
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25 * _SCALE, deadline=None)
 @given(
     n_nodes=st.integers(min_value=1, max_value=8),
     max_parallel=st.integers(min_value=0, max_value=4),
@@ -83,7 +89,7 @@ def test_rolling_window_invariants(n_nodes, max_parallel, max_unavailable_pct,
         assert count <= n_nodes, f"{frm}->{to} fired {count} times for {n_nodes} nodes"
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40 * _SCALE, deadline=None)
 @given(
     total=st.integers(min_value=0, max_value=200),
     pct=st.integers(min_value=0, max_value=100),
@@ -102,7 +108,7 @@ _key = st.text(alphabet="abcz-._/", min_size=1, max_size=10).filter(
 _val = st.text(alphabet="abcz123-", min_size=0, max_size=8)
 
 
-@settings(max_examples=80, deadline=None)
+@settings(max_examples=80 * _SCALE, deadline=None)
 @given(
     reqs=st.lists(
         st.one_of(
@@ -145,7 +151,7 @@ def test_label_selector_roundtrip_matches_composed_predicate(reqs, labels):
     assert sel.matches(labels) == expected, (selector, labels)
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60 * _SCALE, deadline=None)
 @given(junk=st.text(max_size=40))
 def test_label_selector_never_crashes_on_junk(junk):
     from k8s_operator_libs_amd.core.meta import LabelSelector
@@ -183,7 +189,7 @@ def _spec_merge_patch(target, patch):
     return patch
 
 
-@settings(max_examples=120, deadline=None)
+@settings(max_examples=120 * _SCALE, deadline=None)
 @given(target=_json, patch=_json)
 def test_merge_patch_matches_rfc_pseudocode(target, patch):
     import copy as _copy
@@ -195,7 +201,7 @@ def test_merge_patch_matches_rfc_pseudocode(target, patch):
     assert got == expected
 
 
-@settings(max_examples=120, deadline=None)
+@settings(max_examples=120 * _SCALE, deadline=None)
 @given(tree=_json)
 def test_native_deep_copy_matches_stdlib(tree):
     """The C++ _jsonops.deep_copy must agree with copy.deepcopy on arbitrary
