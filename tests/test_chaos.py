@@ -22,7 +22,7 @@ from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
 from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
 
 from builders import PodBuilder
-from simenv import SimDaemonSetController
+from simenv import SimDaemonSetController, SimMaintenanceOperator
 from test_state_manager import policy, setup_cluster
 
 
@@ -61,7 +61,20 @@ def test_converges_despite_random_faults(seed):
     client = FakeClient()
     ds, _ = setup_cluster(client, n_nodes=n_nodes, pod_hash="old", ds_hash="new")
     SimDaemonSetController(client.cluster, ds, current_hash="new")
-    manager = ClusterUpgradeStateManager(client)
+    options = None
+    if rng.randrange(2):  # half the schedules run in requestor mode
+        from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
+        from k8s_operator_libs_amd.upgrade.state_manager import StateOptions
+
+        SimMaintenanceOperator(
+            client.cluster,
+            evict_filter=lambda pod: pod["metadata"].get("labels", {})
+            .get("app") != "amd-gpu-validator",
+        )
+        options = StateOptions(requestor=RequestorOptions(
+            use_maintenance_operator=True, requestor_id="amd.gpu.operator",
+            namespace="default"))
+    manager = ClusterUpgradeStateManager(client, options=options)
     pol = policy(maxParallelUpgrades=rng.randrange(0, 3),
                  maxUnavailable="100%",
                  drainSpec={"enable": bool(rng.randrange(2))})
