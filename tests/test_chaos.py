@@ -57,7 +57,7 @@ def _inject_fault(rng, client, n_nodes):
 @given(seed=st.integers(min_value=0, max_value=10_000))
 def test_converges_despite_random_faults(seed):
     rng = random.Random(seed)
-    n_nodes = rng.randrange(2, 5)
+    n_nodes = rng.randrange(2, 8)
     client = FakeClient()
     ds, _ = setup_cluster(client, n_nodes=n_nodes, pod_hash="old", ds_hash="new")
     SimDaemonSetController(client.cluster, ds, current_hash="new")
