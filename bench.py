@@ -259,7 +259,7 @@ def run_rolling_upgrade_benchmark(
         "maxParallelUpgrades": max_parallel,
         "maxUnavailable": "50%",
         "podDeletion": {"force": False, "deleteEmptyDir": True},
-        "drainSpec": {"enable": True, "timeoutSeconds": 300},
+        "drain": {"enable": True, "timeoutSeconds": 300},
     })
     state_key = util.get_upgrade_state_label_key()
 

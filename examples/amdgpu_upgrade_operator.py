@@ -109,7 +109,7 @@ def run_demo(args):
         "maxParallelUpgrades": args.max_parallel,
         "maxUnavailable": "50%",
         "podDeletion": {"deleteEmptyDir": True},
-        "drainSpec": {"enable": True},
+        "drain": {"enable": True},
     })
 
     # demo validator: mark validation pods ready once their node reaches
@@ -190,7 +190,7 @@ def main(argv=None) -> int:
         "autoUpgrade": True,
         "maxParallelUpgrades": args.max_parallel,
         "podDeletion": {"deleteEmptyDir": True},
-        "drainSpec": {"enable": True},
+        "drain": {"enable": True},
     })
 
     from k8s_operator_libs_amd.core.leaderelection import LeaderElector

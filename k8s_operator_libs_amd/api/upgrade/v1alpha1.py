@@ -26,7 +26,7 @@ import math
 import re
 from typing import Optional, Union
 
-from pydantic import BaseModel, ConfigDict, Field, field_validator
+from pydantic import AliasChoices, BaseModel, ConfigDict, Field, field_validator
 
 _PERCENT_RE = re.compile(r"^(\d+)%$")
 
@@ -77,8 +77,15 @@ class WaitForCompletionSpec(_SpecBase):
 
     # Label selector (string form, e.g. "app=training-job") of pods to wait on.
     pod_selector: str = Field(default="", alias="podSelector")
-    # 0 means wait forever.
-    timeout_seconds: int = Field(default=0, ge=0, alias="timeoutSecond")
+    # 0 means wait forever.  Wire name matches the reference json tag
+    # ``timeoutSeconds`` (upgrade_spec.go:63); the round-1 name
+    # ``timeoutSecond`` is still accepted on input for compatibility.
+    timeout_seconds: int = Field(
+        default=0,
+        ge=0,
+        serialization_alias="timeoutSeconds",
+        validation_alias=AliasChoices("timeoutSeconds", "timeoutSecond", "timeout_seconds"),
+    )
 
 
 class PodDeletionSpec(_SpecBase):
@@ -86,7 +93,14 @@ class PodDeletionSpec(_SpecBase):
     (upgrade_spec.go:67-83)."""
 
     force: bool = Field(default=False)
-    timeout_seconds: int = Field(default=300, ge=0, alias="timeoutSecond")
+    # Wire name ``timeoutSeconds`` per the reference json tag
+    # (upgrade_spec.go:77); round-1's ``timeoutSecond`` accepted on input.
+    timeout_seconds: int = Field(
+        default=300,
+        ge=0,
+        serialization_alias="timeoutSeconds",
+        validation_alias=AliasChoices("timeoutSeconds", "timeoutSecond", "timeout_seconds"),
+    )
     delete_emptydir_data: bool = Field(default=False, alias="deleteEmptyDir")
 
 
@@ -114,7 +128,13 @@ class DriverUpgradePolicySpec(_SpecBase):
         default=None, alias="waitForCompletion"
     )
     pod_deletion: Optional[PodDeletionSpec] = Field(default=None, alias="podDeletion")
-    drain_spec: Optional[DrainSpec] = Field(default=None, alias="drainSpec")
+    # Wire name ``drain`` per the reference json tag (upgrade_spec.go:48);
+    # round-1's ``drainSpec`` accepted on input for compatibility.
+    drain_spec: Optional[DrainSpec] = Field(
+        default=None,
+        serialization_alias="drain",
+        validation_alias=AliasChoices("drain", "drainSpec", "drain_spec"),
+    )
 
     @field_validator("max_unavailable")
     @classmethod

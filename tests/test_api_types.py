@@ -22,26 +22,49 @@ def test_policy_defaults():
     assert p.drain_spec is None
 
 
-def test_policy_wire_aliases():
-    p = DriverUpgradePolicySpec.model_validate(
-        {
-            "autoUpgrade": True,
-            "maxParallelUpgrades": 4,
-            "maxUnavailable": 3,
-            "drainSpec": {"enable": True, "timeoutSeconds": 60},
-            "podDeletion": {"force": True, "deleteEmptyDir": True},
-            "waitForCompletion": {"podSelector": "app=job", "timeoutSecond": 30},
-        }
-    )
+def test_policy_reference_wire_format():
+    """A policy document authored against the reference kubebuilder schema
+    (json tags of upgrade_spec.go:27-110) parses and round-trips byte-
+    compatibly: ``drain`` (spec.go:48) and ``timeoutSeconds``
+    (spec.go:63,77,104) are the wire names."""
+    ref_doc = {
+        "autoUpgrade": True,
+        "maxParallelUpgrades": 4,
+        "maxUnavailable": 3,
+        "drain": {"enable": True, "timeoutSeconds": 60},
+        "podDeletion": {"force": True, "timeoutSeconds": 120, "deleteEmptyDir": True},
+        "waitForCompletion": {"podSelector": "app=job", "timeoutSeconds": 30},
+    }
+    p = DriverUpgradePolicySpec.model_validate(ref_doc)
     assert p.max_parallel_upgrades == 4
     assert p.max_unavailable == 3
     assert p.drain_spec.enable and p.drain_spec.timeout_seconds == 60
     assert p.pod_deletion.force and p.pod_deletion.delete_emptydir_data
+    assert p.pod_deletion.timeout_seconds == 120
     assert p.wait_for_completion.pod_selector == "app=job"
-    # round-trip through the wire shape
-    wire = p.model_dump(by_alias=True)
-    assert wire["maxParallelUpgrades"] == 4
+    assert p.wait_for_completion.timeout_seconds == 30
+    # round-trip: emitted wire shape uses reference names and re-parses equal
+    wire = p.model_dump(by_alias=True, exclude_none=True)
+    assert "drain" in wire and "drainSpec" not in wire
+    assert wire["podDeletion"]["timeoutSeconds"] == 120
+    assert "timeoutSecond" not in wire["waitForCompletion"]
     assert DriverUpgradePolicySpec.model_validate(wire) == p
+
+
+def test_policy_legacy_round1_aliases_still_accepted():
+    """Round-1 wire names (``drainSpec``, ``timeoutSecond``) remain valid
+    on input so existing in-repo documents keep parsing."""
+    p = DriverUpgradePolicySpec.model_validate(
+        {
+            "drainSpec": {"enable": True, "timeoutSeconds": 60},
+            "waitForCompletion": {"podSelector": "app=job", "timeoutSecond": 30},
+        }
+    )
+    assert p.drain_spec.enable and p.drain_spec.timeout_seconds == 60
+    assert p.wait_for_completion.timeout_seconds == 30
+    # but output is normalized to the reference format
+    wire = p.model_dump(by_alias=True, exclude_none=True)
+    assert "drain" in wire and "drainSpec" not in wire
 
 
 def test_nested_defaults_match_reference():
@@ -97,8 +120,8 @@ def test_openapi_v3_schema_for_crd_embedding():
     assert props["maxParallelUpgrades"]["minimum"] == 0
     assert props["maxUnavailable"]["x-kubernetes-int-or-string"] is True
     assert props["maxUnavailable"]["default"] == "25%"
-    assert props["drainSpec"]["properties"]["timeoutSeconds"]["default"] == 300
-    assert props["podDeletion"]["properties"]["timeoutSecond"]["default"] == 300
+    assert props["drain"]["properties"]["timeoutSeconds"]["default"] == 300
+    assert props["podDeletion"]["properties"]["timeoutSeconds"]["default"] == 300
     # structural: no $refs/anyOf remain anywhere
     import json
     text = json.dumps(schema)
