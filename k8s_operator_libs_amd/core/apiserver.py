@@ -84,39 +84,103 @@ def create_app(cluster: Optional[FakeCluster] = None):
 
     async def _list(group, version, plural, request: Request, namespace: str = ""):
         if request.query_params.get("watch") in ("true", "1"):
-            return _watch_stream(group, version, plural)
+            return _watch_stream(group, version, plural, request, namespace)
 
         def run():
             api_version, kind = _resolve(group, version, plural)
-            items = cluster.list(
+            items, rv = cluster.list_with_meta(
                 api_version, kind,
                 namespace=namespace or None,
                 label_selector=request.query_params.get("labelSelector", ""),
                 field_selector=request.query_params.get("fieldSelector", ""),
             )
-            return {"kind": f"{kind}List", "apiVersion": api_version, "items": items}
+            return {
+                "kind": f"{kind}List",
+                "apiVersion": api_version,
+                "metadata": {"resourceVersion": rv},
+                "items": items,
+            }
         return _handle(run)
 
-    def _watch_stream(group, version, plural):
-        """Kubernetes-style watch: stream newline-delimited WatchEvent JSON."""
+    def _watch_stream(group, version, plural, request: Request, namespace: str = ""):
+        """Kubernetes watch protocol: newline-delimited WatchEvent JSON.
+
+        Semantics match a real apiserver (no custom handshake frames):
+
+        - ``resourceVersion`` anchors the stream; an expired RV is signalled
+          as an in-stream ``ERROR`` event carrying a 410 Status (the wire
+          shape client-go expects), after which the stream closes;
+        - ``allowWatchBookmarks=true`` opts into periodic ``BOOKMARK``
+          events whose object carries only ``metadata.resourceVersion``;
+        - ``timeoutSeconds`` closes the stream server-side when it elapses;
+        - ``labelSelector`` filters events (stops-matching -> DELETED);
+        - namespace-scoped paths watch only that namespace.
+        """
         from fastapi.responses import StreamingResponse
 
+        from .errors import GoneError
+
+        qp = request.query_params
         try:
             api_version, kind = _resolve(group, version, plural)
         except ApiError as exc:
             return JSONResponse(_status_body(exc), status_code=exc.code)
-        watch = cluster.watch(api_version, kind)
+
+        resource_version = qp.get("resourceVersion") or None
+        label_selector = qp.get("labelSelector", "")
+        bookmarks = qp.get("allowWatchBookmarks") in ("true", "1")
+        try:
+            timeout_s = float(qp.get("timeoutSeconds", "0")) or None
+        except ValueError:
+            timeout_s = None
+
+        try:
+            watch = cluster.watch(
+                api_version, kind,
+                namespace=namespace or None,
+                resource_version=resource_version,
+                label_selector=label_selector,
+            )
+        except GoneError as exc:
+            # expired before the stream even opened: stream a single ERROR
+            # event (a real apiserver answers 200 + in-stream ERROR Status)
+            gone_message = exc.message
+
+            def gone_gen():
+                yield json.dumps({
+                    "type": "ERROR",
+                    "object": {
+                        "kind": "Status", "apiVersion": "v1",
+                        "status": "Failure", "reason": "Expired",
+                        "message": gone_message, "code": 410,
+                    },
+                }) + "\n"
+            return StreamingResponse(gone_gen(), media_type="application/json")
+        except ApiError as exc:
+            return JSONResponse(_status_body(exc), status_code=exc.code)
 
         def gen():
+            deadline = time.monotonic() + timeout_s if timeout_s else None
+            last_bookmark = None
             try:
-                # immediate bookmark: the client knows the watch is
-                # registered server-side before it issues its initial LIST,
-                # closing the lost-event gap (real apiservers close it with
-                # resourceVersion-anchored watches)
-                yield json.dumps({"type": "BOOKMARK", "object": None}) + "\n"
                 while True:
+                    if deadline is not None and time.monotonic() >= deadline:
+                        return
                     item = watch.next(timeout=0.5)
                     if item is None:
+                        if bookmarks:
+                            rv = watch.bookmark_rv()
+                            if rv is not None and rv != last_bookmark:
+                                last_bookmark = rv
+                                yield json.dumps({
+                                    "type": "BOOKMARK",
+                                    "object": {
+                                        "kind": kind,
+                                        "apiVersion": api_version,
+                                        "metadata": {"resourceVersion": rv},
+                                    },
+                                }) + "\n"
+                                continue
                         yield ""  # keep-alive; also surfaces disconnects
                         continue
                     event_type, obj = item

@@ -31,6 +31,10 @@ from .meta import K8sObject
 
 
 class _Informer:
+    #: reconnect backoff bounds (seconds) — client-go reflector-style
+    BACKOFF_BASE = 0.1
+    BACKOFF_MAX = 5.0
+
     def __init__(self, delegate: Client, api_version: str, kind: str,
                  sync_delay: float = 0.0) -> None:
         self.api_version = api_version
@@ -43,64 +47,118 @@ class _Informer:
         self._stop = threading.Event()
         self._watch = None
         self._thread: Optional[threading.Thread] = None
+        # last resourceVersion this informer has processed: the anchor for
+        # lossless watch reconnects (client-go reflector lastSyncResourceVersion)
+        self._last_rv: Optional[str] = None
 
     def start(self) -> None:
-        self._watch = self._delegate.watch(self.api_version, self.kind)
         self._thread = threading.Thread(target=self._run, daemon=True)
         self._thread.start()
 
     def _relist(self) -> None:
-        """Full LIST replacing the store (initial sync and watch-reconnect
-        recovery — a dropped stream may have lost events)."""
-        objs = self._delegate.list(self.api_version, self.kind)
+        """Full LIST replacing the store and re-anchoring ``_last_rv``
+        (initial sync and 410-Gone recovery)."""
+        objs, rv = self._delegate.list_with_meta(self.api_version, self.kind)
         with self._lock:
             self._store = {
                 (meta.namespace(obj), meta.name(obj)): obj for obj in objs
             }
+            self._last_rv = rv
+
+    def _open_watch(self):
+        """WATCH anchored at the last processed resourceVersion; a delegate
+        without RV support (``_last_rv`` None) gets a live-only watch, in
+        which case the caller must LIST *after* opening to close the gap."""
+        return self._delegate.watch(
+            self.api_version, self.kind, resource_version=self._last_rv
+        )
 
     def _run(self) -> None:
-        # initial LIST (after the watch opened, so no events are lost);
-        # retried so a briefly-unavailable apiserver can't kill the informer
+        from .errors import GoneError
+
+        backoff = self.BACKOFF_BASE
+        # reflector bootstrap: LIST (capture list RV) -> WATCH(from that RV).
+        # With an RV-anchored delegate nothing can be lost in between; with a
+        # live-only delegate, watch-then-relist covers the gap instead.
         while not self._stop.is_set():
             try:
                 self._relist()
+                if self._last_rv is None:
+                    # legacy live-only delegate: open watch first, then
+                    # relist so events during the LIST aren't dropped
+                    self._watch = self._open_watch()
+                    self._relist()
+                else:
+                    self._watch = self._open_watch()
                 break
             except Exception:
-                self._stop.wait(0.5)
+                self._stop.wait(backoff)
+                backoff = min(backoff * 2, self.BACKOFF_MAX)
         self._synced.set()
+        backoff = self.BACKOFF_BASE
         while not self._stop.is_set():
             item = self._watch.next(timeout=0.2)
             if item is None:
                 alive = getattr(self._watch, "alive", None)
                 if alive is not None and not alive():
-                    # stream dropped: reconnect, then relist to recover any
-                    # events lost in the gap (informer resync semantics)
+                    # stream dropped: re-watch from the last processed RV —
+                    # only a 410 (resume window expired) forces a full relist
                     try:
-                        self._watch = self._delegate.watch(self.api_version, self.kind)
-                        self._relist()
+                        self._watch = self._open_watch()
+                        if self._last_rv is None:
+                            self._relist()
+                        backoff = self.BACKOFF_BASE
+                    except GoneError:
+                        try:
+                            self._relist()
+                            self._watch = self._open_watch()
+                            backoff = self.BACKOFF_BASE
+                        except Exception:
+                            self._stop.wait(backoff)
+                            backoff = min(backoff * 2, self.BACKOFF_MAX)
                     except Exception:
-                        self._stop.wait(1.0)
+                        self._stop.wait(backoff)
+                        backoff = min(backoff * 2, self.BACKOFF_MAX)
+                continue
+            event_type, obj = item
+            if event_type == "BOOKMARK":
+                rv = (obj or {}).get("metadata", {}).get("resourceVersion")
+                if rv:
+                    self._last_rv = rv
+                continue
+            if event_type == "ERROR":
+                # Kubernetes signals an expired watch as an in-stream ERROR
+                # Status (code 410): relist + re-watch from the fresh RV
+                code = (obj or {}).get("code")
+                try:
+                    if code == 410:
+                        self._relist()
+                    self._watch.stop()
+                    self._watch = self._open_watch()
+                except Exception:
+                    self._stop.wait(backoff)
+                    backoff = min(backoff * 2, self.BACKOFF_MAX)
                 continue
             if self._sync_delay:
                 time.sleep(self._sync_delay)
-            event_type, obj = item
             key = (meta.namespace(obj), meta.name(obj))
             with self._lock:
+                rv = meta.resource_version(obj)
                 if event_type == "DELETED":
                     self._store.pop(key, None)
+                    self._last_rv = rv or self._last_rv
                 else:
                     current = self._store.get(key)
                     # resourceVersions are monotonic ints in this stack;
                     # never regress the cache on out-of-order delivery
                     if current is not None:
                         try:
-                            if int(meta.resource_version(obj)) < int(
-                                meta.resource_version(current)
-                            ):
+                            if int(rv) < int(meta.resource_version(current)):
                                 continue
                         except ValueError:
                             pass
                     self._store[key] = obj
+                    self._last_rv = rv or self._last_rv
 
     def stop(self) -> None:
         self._stop.set()
@@ -201,5 +259,16 @@ class CachedClient(Client):
     def evict_pod(self, name, namespace):
         self._delegate.evict_pod(name, namespace)
 
-    def watch(self, api_version, kind):
-        return self._delegate.watch(api_version, kind)
+    def watch(self, api_version, kind, namespace=None, resource_version=None,
+              label_selector=""):
+        return self._delegate.watch(
+            api_version, kind, namespace=namespace,
+            resource_version=resource_version, label_selector=label_selector,
+        )
+
+    def list_with_meta(self, api_version, kind, namespace=None,
+                       label_selector="", field_selector=""):
+        return self._delegate.list_with_meta(
+            api_version, kind, namespace=namespace,
+            label_selector=label_selector, field_selector=field_selector,
+        )

@@ -49,8 +49,38 @@ class Client(abc.ABC):
     @abc.abstractmethod
     def evict_pod(self, name: str, namespace: str) -> None: ...
 
-    def watch(self, api_version: str, kind: str) -> Watch:
+    def watch(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        resource_version: Optional[str] = None,
+        label_selector: str = "",
+    ) -> Watch:
+        """Open a watch stream.  ``resource_version`` anchors the stream per
+        Kubernetes semantics (None = live-only, "0" = synthetic ADDEDs then
+        live, otherwise replay-after-RV with 410
+        :class:`~k8s_operator_libs_amd.core.errors.GoneError` when the
+        resume window has expired)."""
         raise NotImplementedError("this client does not support watches")
+
+    def list_with_meta(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: str = "",
+        field_selector: str = "",
+    ):
+        """LIST returning ``(items, list_resource_version)`` — the anchor for
+        a lossless LIST-then-WATCH(rv) reflector loop.  Clients without
+        list-level RV support return ``(items, None)`` and the informer falls
+        back to live-only watches."""
+        return (
+            self.list(api_version, kind, namespace=namespace,
+                      label_selector=label_selector, field_selector=field_selector),
+            None,
+        )
 
     def patch_status(
         self, api_version: str, kind: str, name: str, status: dict, namespace: str = ""
@@ -129,5 +159,16 @@ class FakeClient(Client):
     def evict_pod(self, name, namespace):
         self.cluster.evict_pod(name, namespace)
 
-    def watch(self, api_version, kind):
-        return self.cluster.watch(api_version, kind)
+    def watch(self, api_version, kind, namespace=None, resource_version=None,
+              label_selector=""):
+        return self.cluster.watch(
+            api_version, kind, namespace=namespace,
+            resource_version=resource_version, label_selector=label_selector,
+        )
+
+    def list_with_meta(self, api_version, kind, namespace=None,
+                       label_selector="", field_selector=""):
+        return self.cluster.list_with_meta(
+            api_version, kind, namespace=namespace,
+            label_selector=label_selector, field_selector=field_selector,
+        )

@@ -31,6 +31,13 @@ class BadRequestError(ApiError):
     code = 400
 
 
+class GoneError(ApiError):
+    """Watch resume window expired (HTTP 410, status reason ``Expired``) —
+    the client must relist and re-watch from the fresh resourceVersion."""
+
+    code = 410
+
+
 def is_not_found(exc: BaseException) -> bool:
     return isinstance(exc, NotFoundError)
 

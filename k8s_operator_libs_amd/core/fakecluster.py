@@ -14,7 +14,7 @@ which serves it over HTTP so the REST client can be tested wire-level.
 
 from __future__ import annotations
 
-import itertools
+import collections
 import logging
 import queue
 import threading
@@ -54,19 +54,76 @@ _BUILTIN_KINDS: Dict[Tuple[str, str], Tuple[str, bool]] = {
 
 class Watch:
     """A watch stream: a queue of ``(event_type, object)`` tuples where
-    event_type is ``ADDED`` / ``MODIFIED`` / ``DELETED``."""
+    event_type is ``ADDED`` / ``MODIFIED`` / ``DELETED`` (or ``ERROR`` with a
+    Status object — the Kubernetes wire shape for an expired watch).
 
-    def __init__(self, cluster: "FakeCluster", key: Tuple[str, str]) -> None:
+    Supports the real apiserver's filtering semantics: an optional namespace
+    scope and label selector, where an object that stops matching the
+    selector is delivered as ``DELETED`` and one that starts matching as
+    ``ADDED`` (client-go reflector contract)."""
+
+    def __init__(
+        self,
+        cluster: "FakeCluster",
+        key: Tuple[str, str],
+        namespace: Optional[str] = None,
+        label_selector: str = "",
+    ) -> None:
         self._cluster = cluster
         self._key = key
+        self._namespace = namespace or None
+        self._selector = (
+            meta.parse_label_selector(label_selector) if label_selector else None
+        )
+        # object keys currently matching the selector (for the
+        # stops-matching -> DELETED transform); only used with a selector
+        self._matched: set = set()
         self.events: "queue.Queue[Tuple[str, K8sObject]]" = queue.Queue()
         self._stopped = False
+
+    def _deliver(self, event_type: str, snapshot: K8sObject) -> None:
+        """Apply namespace/selector filtering; called under the cluster lock
+        so delivery order matches resourceVersion order."""
+        if self._namespace is not None and meta.namespace(snapshot) != self._namespace:
+            return
+        if self._selector is not None:
+            okey = (meta.namespace(snapshot), meta.name(snapshot))
+            matches = event_type != "DELETED" and self._selector.matches_object(snapshot)
+            was_matched = okey in self._matched
+            if event_type == "DELETED":
+                # deliver if this watch saw the object OR the final snapshot
+                # matches (a resumed watch replaying a delete it never saw
+                # the ADDED for must still get the DELETED)
+                if not was_matched and not self._selector.matches_object(snapshot):
+                    return
+                self._matched.discard(okey)
+            elif matches and not was_matched:
+                self._matched.add(okey)
+                event_type = "ADDED"
+            elif matches:
+                pass  # stays MODIFIED
+            elif was_matched:
+                self._matched.discard(okey)
+                event_type = "DELETED"
+            else:
+                return
+        self.events.put((event_type, snapshot))
 
     def next(self, timeout: Optional[float] = None) -> Optional[Tuple[str, K8sObject]]:
         try:
             return self.events.get(timeout=timeout)
         except queue.Empty:
             return None
+
+    def bookmark_rv(self) -> Optional[str]:
+        """Current cluster resourceVersion, but only when this watch's queue
+        is drained — taken under the cluster lock so no event at or below the
+        returned RV can still be pending delivery.  This makes the value safe
+        to hand out as a BOOKMARK a client may resume from."""
+        with self._cluster._lock:
+            if not self.events.empty():
+                return None
+            return str(self._cluster._rv_counter)
 
     def stop(self) -> None:
         self._stopped = True
@@ -79,14 +136,26 @@ class Watch:
 class FakeCluster:
     """Thread-safe in-memory Kubernetes object store."""
 
+    #: events retained per kind for resourceVersion-anchored watch resume;
+    #: resuming below the window -> 410 Gone (matches the apiserver's
+    #: bounded etcd watch cache)
+    WATCH_HISTORY = 2048
+
     def __init__(self) -> None:
         self._lock = threading.RLock()
         # (apiVersion, kind) -> {(namespace, name): object}
         self._store: Dict[Tuple[str, str], Dict[Tuple[str, str], K8sObject]] = {}
-        self._rv = itertools.count(1)
+        self._rv_counter = 0
         self._kinds = dict(_BUILTIN_KINDS)
         self._watches: Dict[Tuple[str, str], List[Watch]] = {}
         self._change_hooks: List[Callable[[str, K8sObject], None]] = []
+        # Watch-resume cache: per kind, a bounded deque of
+        # (rv_int, event_type, snapshot).  Recording starts when the first
+        # watch on the kind opens (before that, any RV-anchored resume gets
+        # 410 Gone, which is the correct "too old" answer).
+        self._history: Dict[Tuple[str, str], "collections.deque"] = {}
+        # RV from which each kind's history is complete
+        self._history_start: Dict[Tuple[str, str], int] = {}
         # spec.nodeName index for pods: nodeName -> {(ns, name)}; keeps
         # per-node pod LISTs (the reconcile loop's hottest query) O(pods on
         # node) instead of O(all pods)
@@ -127,7 +196,14 @@ class FakeCluster:
     # -- internal helpers ----------------------------------------------------
 
     def _next_rv(self) -> str:
-        return str(next(self._rv))
+        with self._lock:
+            self._rv_counter += 1
+            return str(self._rv_counter)
+
+    def current_rv(self) -> str:
+        """The cluster-level resourceVersion a LIST would carry right now."""
+        with self._lock:
+            return str(self._rv_counter)
 
     @staticmethod
     def _obj_key(obj: K8sObject) -> Tuple[str, str]:
@@ -140,11 +216,19 @@ class FakeCluster:
     def _notify(self, event_type: str, obj: K8sObject) -> None:
         key = self._obj_key(obj)
         watches = self._watches.get(key)
-        if not watches and not self._change_hooks:
+        history = self._history.get(key)
+        if not watches and not self._change_hooks and history is None:
             return  # nobody listening: skip the snapshot copy (hot path)
         snapshot = meta.deep_copy(obj)
+        if history is not None:
+            try:
+                history.append(
+                    (int(meta.resource_version(snapshot)), event_type, snapshot)
+                )
+            except ValueError:
+                pass
         for w in watches or ():
-            w.events.put((event_type, snapshot))
+            w._deliver(event_type, snapshot)
         for hook in self._change_hooks:
             # hooks are observers (simulated controllers): a failing hook
             # must never fail the API mutation that triggered it
@@ -258,6 +342,23 @@ class FakeCluster:
             out.sort(key=lambda o: (meta.namespace(o), meta.name(o)))
             return out
 
+    def list_with_meta(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: str = "",
+        field_selector: str = "",
+    ) -> Tuple[List[K8sObject], str]:
+        """LIST plus the list-level ``metadata.resourceVersion`` a watch can
+        be anchored at (the reflector's LIST-then-WATCH(rv) contract)."""
+        with self._lock:
+            items = self.list(
+                api_version, kind, namespace=namespace,
+                label_selector=label_selector, field_selector=field_selector,
+            )
+            return items, str(self._rv_counter)
+
     def update(self, obj: K8sObject) -> K8sObject:
         obj = meta.deep_copy(obj)
         api_version, kind = self._obj_key(obj)
@@ -347,6 +448,9 @@ class FakeCluster:
                 return
             self._index_pod(stored, remove=True)
             del bucket[(ns, name)]
+            # deletion gets its own resourceVersion (real apiservers do this;
+            # required so an RV-anchored watch resume replays the delete)
+            stored["metadata"]["resourceVersion"] = self._next_rv()
             self._notify("DELETED", stored)
 
     def _finalize_if_ready(self, api_version: str, kind: str, ns: str, name: str) -> bool:
@@ -360,6 +464,7 @@ class FakeCluster:
         if "deletionTimestamp" in md and not (md.get("finalizers") or []):
             self._index_pod(stored, remove=True)
             del bucket[(ns, name)]
+            stored["metadata"]["resourceVersion"] = self._next_rv()
             self._notify("DELETED", stored)
             return True
         return False
@@ -427,11 +532,69 @@ class FakeCluster:
 
     # -- watches -------------------------------------------------------------
 
-    def watch(self, api_version: str, kind: str) -> Watch:
+    def _history_floor(self, key: Tuple[str, str]) -> Optional[int]:
+        """Lowest RV from which this kind's event history is complete, or
+        None if history has never been recorded."""
+        start = self._history_start.get(key)
+        if start is None:
+            return None
+        hist = self._history[key]
+        if len(hist) == hist.maxlen:
+            # ring evicted events: complete only from the oldest retained - 1
+            return hist[0][0] - 1
+        return start
+
+    def watch(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        resource_version: Optional[str] = None,
+        label_selector: str = "",
+    ) -> Watch:
+        """Open a watch stream with Kubernetes resourceVersion semantics:
+
+        - ``resource_version=None`` — live events from now on (the legacy
+          in-process behavior; real apiservers treat an unset RV as "list
+          current state then watch", which :meth:`list_with_meta` + an
+          anchored watch composes for the informer).
+        - ``"0"`` — synthetic ``ADDED`` for every currently-stored matching
+          object, then live events.
+        - any other value — replay retained history with RV strictly greater
+          than the given value, then live; raises
+          :class:`~k8s_operator_libs_amd.core.errors.GoneError` (410) when
+          the requested RV has fallen out of the bounded history window.
+        """
+        from .errors import GoneError
+
         with self._lock:
             self.lookup_kind(api_version, kind)
-            w = Watch(self, (api_version, kind))
-            self._watches.setdefault((api_version, kind), []).append(w)
+            key = (api_version, kind)
+            # first watch on a kind turns its resume history on
+            if key not in self._history:
+                self._history[key] = collections.deque(maxlen=self.WATCH_HISTORY)
+                self._history_start[key] = self._rv_counter
+            w = Watch(self, key, namespace=namespace, label_selector=label_selector)
+            if resource_version == "0":
+                for obj in self._store.get(key, {}).values():
+                    w._deliver("ADDED", meta.deep_copy(obj))
+            elif resource_version not in (None, ""):
+                try:
+                    rv_int = int(resource_version)
+                except ValueError:
+                    raise BadRequestError(
+                        f"invalid resourceVersion {resource_version!r}"
+                    ) from None
+                floor = self._history_floor(key)
+                if floor is None or rv_int < floor:
+                    raise GoneError(
+                        f"too old resource version: {resource_version} "
+                        f"(history starts at {floor})"
+                    )
+                for ev_rv, ev_type, snapshot in self._history[key]:
+                    if ev_rv > rv_int:
+                        w._deliver(ev_type, snapshot)
+            self._watches.setdefault(key, []).append(w)
             return w
 
     # -- CRD establishment ---------------------------------------------------

@@ -51,8 +51,18 @@ _KIND_INFO = {
 
 
 def _lower_plural(kind: str) -> str:
+    """English pluralization matching apimachinery's namer conventions:
+    ``NetworkPolicy`` -> ``networkpolicies``, ``Ingress`` -> ``ingresses``,
+    ``Endpoints`` -> ``endpoints``.  Only a fallback — registered kinds and
+    discovery (see :meth:`RestClient._kind_info`) take precedence."""
     k = kind.lower()
-    return k + ("es" if k.endswith("s") else "s")
+    if k.endswith(("ss", "x", "z", "ch", "sh")):
+        return k + "es"
+    if k.endswith("s"):
+        return k  # already plural-looking (Endpoints)
+    if k.endswith("y") and len(k) > 1 and k[-2] not in "aeiou":
+        return k[:-1] + "ies"
+    return k + "s"
 
 
 class RestClient(Client):
@@ -145,8 +155,28 @@ class RestClient(Client):
     def _kind_info(self, api_version: str, kind: str) -> tuple:
         info = self._kinds.get((api_version, kind))
         if info is None:
-            info = (_lower_plural(kind), True)
+            # resolve through API discovery (authoritative, like client-go's
+            # RESTMapper) before falling back to English pluralization rules
+            info = self._discover_kind(api_version, kind) or (_lower_plural(kind), True)
+            self._kinds[(api_version, kind)] = info
         return info
+
+    def _discover_kind(self, api_version: str, kind: str) -> Optional[tuple]:
+        prefix = f"/api/{api_version}" if "/" not in api_version else f"/apis/{api_version}"
+        try:
+            resp = self._request("GET", prefix)
+        except ApiError:
+            return None
+        if resp.status_code != 200:
+            return None
+        try:
+            resources = resp.json().get("resources", [])
+        except ValueError:
+            return None
+        for res in resources:
+            if res.get("kind") == kind and "/" not in res.get("name", ""):
+                return (res["name"], bool(res.get("namespaced", True)))
+        return None
 
     def _collection_path(self, api_version: str, kind: str, namespace: str) -> str:
         plural, namespaced = self._kind_info(api_version, kind)
@@ -191,6 +221,13 @@ class RestClient(Client):
         return resp.json()
 
     def list(self, api_version, kind, namespace=None, label_selector="", field_selector=""):
+        return self.list_with_meta(
+            api_version, kind, namespace=namespace,
+            label_selector=label_selector, field_selector=field_selector,
+        )[0]
+
+    def list_with_meta(self, api_version, kind, namespace=None,
+                       label_selector="", field_selector=""):
         params = {}
         if label_selector:
             params["labelSelector"] = label_selector
@@ -199,7 +236,9 @@ class RestClient(Client):
         path = self._collection_path(api_version, kind, namespace or "")
         resp = self._request("GET", path, params=params)
         self._raise_for(resp)
-        return resp.json().get("items", [])
+        body = resp.json()
+        rv = (body.get("metadata") or {}).get("resourceVersion")
+        return body.get("items", []), rv
 
     def create(self, obj):
         api_version, kind = meta.api_version(obj), meta.kind(obj)
@@ -247,12 +286,20 @@ class RestClient(Client):
         resp = self._request("POST", path, content=json.dumps(body))
         self._raise_for(resp)
 
-    def watch(self, api_version: str, kind: str):
-        """Open a Kubernetes-style watch stream (``?watch=true``), returning
-        an object with ``next(timeout)`` / ``stop()`` like FakeCluster's
-        Watch — so :class:`~k8s_operator_libs_amd.core.cache.CachedClient`
-        runs unchanged over REST."""
-        return _HttpWatch(self, api_version, kind)
+    def watch(self, api_version: str, kind: str, namespace=None,
+              resource_version=None, label_selector=""):
+        """Open a Kubernetes watch stream (``?watch=true``) with standard
+        query parameters: ``resourceVersion`` anchoring, ``labelSelector``
+        filtering, namespace-scoped paths, and ``allowWatchBookmarks`` so
+        the server can advance the resume point while idle.  Returns an
+        object with ``next(timeout)`` / ``stop()`` like FakeCluster's Watch —
+        :class:`~k8s_operator_libs_amd.core.cache.CachedClient` runs
+        unchanged over REST.  An expired resume point surfaces as a
+        410 ``GoneError`` (HTTP status) or an in-stream ``("ERROR", status)``
+        event, both of which the informer answers with relist."""
+        return _HttpWatch(self, api_version, kind, namespace=namespace,
+                          resource_version=resource_version,
+                          label_selector=label_selector)
 
     # -- discovery (for crdutil.wait_for_crds) ----------------------------------
 
@@ -275,25 +322,49 @@ class RestClient(Client):
 
 class _HttpWatch:
     """Streaming watch over HTTP: a reader thread feeds a queue of
-    ``(event_type, object)`` tuples parsed from newline-delimited
-    WatchEvent JSON."""
+    ``(event_type, object)`` tuples parsed from newline-delimited WatchEvent
+    JSON — the real apiserver framing, with no custom handshake.  The
+    constructor returns once the response headers arrive (the server has
+    registered the watch by then); against an RV-anchored server nothing can
+    be lost before that because the informer resumes from its list RV.
+    ``ERROR`` and ``BOOKMARK`` events pass through to the consumer."""
 
-    def __init__(self, client: RestClient, api_version: str, kind: str) -> None:
+    def __init__(self, client: RestClient, api_version: str, kind: str,
+                 namespace=None, resource_version=None,
+                 label_selector="") -> None:
         import queue
         import threading
 
         self._queue: "queue.Queue" = queue.Queue()
         self._stop = threading.Event()
         self._connected = threading.Event()
-        path = client._collection_path(api_version, kind, "")
+        path = client._collection_path(api_version, kind, namespace or "")
+        params = {"watch": "true", "allowWatchBookmarks": "true"}
+        if resource_version is not None:
+            params["resourceVersion"] = str(resource_version)
+        if label_selector:
+            params["labelSelector"] = label_selector
+        self._error: Optional[BaseException] = None
 
         def reader():
             try:
                 # stream through the client's own session so TLS verification,
                 # auth headers and base_url apply to watches too
                 with client._http.stream(
-                    "GET", path, params={"watch": "true"}, timeout=None,
+                    "GET", path, params=params, timeout=None,
                 ) as resp:
+                    if resp.status_code == 410:
+                        from .errors import GoneError
+
+                        self._error = GoneError("watch resume point expired")
+                        return
+                    if resp.status_code >= 400:
+                        err = ApiError(f"watch rejected: HTTP {resp.status_code}")
+                        err.code = resp.status_code
+                        self._error = err
+                        return
+                    # headers received: the server has registered the watch
+                    self._connected.set()
                     for line in resp.iter_lines():
                         if self._stop.is_set():
                             break
@@ -304,14 +375,11 @@ class _HttpWatch:
                             event = json.loads(line)
                         except ValueError:
                             continue
-                        if event.get("type") == "BOOKMARK":
-                            self._connected.set()
-                            continue
                         self._queue.put((event.get("type"), event.get("object")))
             except Exception as exc:
                 if not self._stop.is_set():
                     # stream dropped: alive() goes False and the informer
-                    # layer reconnects + relists; no point crashing a thread
+                    # layer reconnects from its last RV; no point crashing
                     import logging as _logging
 
                     _logging.getLogger(__name__).warning(
@@ -322,10 +390,9 @@ class _HttpWatch:
 
         self._thread = threading.Thread(target=reader, daemon=True)
         self._thread.start()
-        # Block until the server has registered the watch: events fired
-        # after this point (e.g. by an informer's post-watch initial LIST)
-        # cannot be lost.
         self._connected.wait(10.0)
+        if self._error is not None:
+            raise self._error
 
     def next(self, timeout: Optional[float] = None):
         import queue
