@@ -5,7 +5,7 @@
 PY ?= python3
 HIPCC ?= /opt/rocm/bin/hipcc
 
-.PHONY: all build test test-par test-gpu bench demo lint coverage clean
+.PHONY: all build test test-par test-gpu test-real-apiserver bench demo lint coverage clean
 
 all: build test
 
@@ -26,6 +26,22 @@ test-par:
 # GPU test suite: requires an MI355X with ROCm.
 test-gpu:
 	$(PY) -m pytest tests/ -q -m gpu --timeout 600
+
+# Conformance surface against a REAL kube-apiserver + etcd (the reference's
+# envtest strategy, upgrade_suit_test.go:86-93 / Makefile:76-78).  Binaries
+# are discovered via $$KUBEBUILDER_ASSETS / $$TEST_ASSET_KUBE_APISERVER +
+# $$TEST_ASSET_ETCD / /usr/local/kubebuilder/bin / $$PATH; alternatively
+# point $$CONFORMANCE_URL (+$$CONFORMANCE_TOKEN) at a disposable cluster
+# (e.g. kind).  Fails loudly when no real substrate is available so it can't
+# silently pass on the mini-apiserver alone — see docs/testing.md.
+test-real-apiserver:
+	@$(PY) -c "from k8s_operator_libs_amd.testing.envtest import find_assets; \
+	import os, sys; \
+	ok = find_assets() is not None or bool(os.environ.get('CONFORMANCE_URL')); \
+	sys.exit(0 if ok else (print('ERROR: no kube-apiserver/etcd binaries found and CONFORMANCE_URL unset.', \
+	'Install envtest assets (setup-envtest use -p path) and export KUBEBUILDER_ASSETS,', \
+	'or point CONFORMANCE_URL at a disposable cluster.', file=sys.stderr) or 1))"
+	$(PY) -m pytest tests/test_conformance.py -q --timeout 600
 
 # Flagship benchmark (BASELINE config #3).
 bench:

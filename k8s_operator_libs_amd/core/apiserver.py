@@ -88,16 +88,26 @@ def create_app(cluster: Optional[FakeCluster] = None):
 
         def run():
             api_version, kind = _resolve(group, version, plural)
-            items, rv = cluster.list_with_meta(
+            qp = request.query_params
+            try:
+                limit = int(qp.get("limit", "0"))
+            except ValueError:
+                limit = 0
+            items, rv, next_token = cluster.list_paged(
                 api_version, kind,
                 namespace=namespace or None,
-                label_selector=request.query_params.get("labelSelector", ""),
-                field_selector=request.query_params.get("fieldSelector", ""),
+                label_selector=qp.get("labelSelector", ""),
+                field_selector=qp.get("fieldSelector", ""),
+                limit=limit,
+                continue_token=qp.get("continue", ""),
             )
+            md = {"resourceVersion": rv}
+            if next_token:
+                md["continue"] = next_token
             return {
                 "kind": f"{kind}List",
                 "apiVersion": api_version,
-                "metadata": {"resourceVersion": rv},
+                "metadata": md,
                 "items": items,
             }
         return _handle(run)

@@ -253,8 +253,12 @@ class CachedClient(Client):
     def patch(self, api_version, kind, name, patch, namespace=""):
         return self._delegate.patch(api_version, kind, name, patch, namespace)
 
-    def delete(self, api_version, kind, name, namespace=""):
-        self._delegate.delete(api_version, kind, name, namespace)
+    def delete(self, api_version, kind, name, namespace="", grace_period_seconds=None):
+        try:
+            self._delegate.delete(api_version, kind, name, namespace,
+                                  grace_period_seconds=grace_period_seconds)
+        except TypeError:
+            self._delegate.delete(api_version, kind, name, namespace)
 
     def evict_pod(self, name, namespace):
         self._delegate.evict_pod(name, namespace)

@@ -44,7 +44,10 @@ class Client(abc.ABC):
     ) -> K8sObject: ...
 
     @abc.abstractmethod
-    def delete(self, api_version: str, kind: str, name: str, namespace: str = "") -> None: ...
+    def delete(self, api_version: str, kind: str, name: str, namespace: str = "") -> None:
+        """Delete an object.  Implementations may accept a
+        ``grace_period_seconds`` keyword (pod graceful-deletion override);
+        in-memory substrates delete immediately and ignore it."""
 
     @abc.abstractmethod
     def evict_pod(self, name: str, namespace: str) -> None: ...
@@ -153,7 +156,8 @@ class FakeClient(Client):
     def patch(self, api_version, kind, name, patch, namespace=""):
         return self.cluster.patch(api_version, kind, name, patch, namespace)
 
-    def delete(self, api_version, kind, name, namespace=""):
+    def delete(self, api_version, kind, name, namespace="", grace_period_seconds=None):
+        # in-memory deletion is immediate; grace period has no kubelet to honour
         self.cluster.delete(api_version, kind, name, namespace)
 
     def evict_pod(self, name, namespace):

@@ -15,6 +15,7 @@ which serves it over HTTP so the REST client can be tested wire-level.
 from __future__ import annotations
 
 import collections
+import json
 import logging
 import queue
 import threading
@@ -358,6 +359,57 @@ class FakeCluster:
                 label_selector=label_selector, field_selector=field_selector,
             )
             return items, str(self._rv_counter)
+
+    def list_paged(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: str = "",
+        field_selector: str = "",
+        limit: int = 0,
+        continue_token: str = "",
+    ) -> Tuple[List[K8sObject], str, str]:
+        """Chunked LIST (``limit``/``continue`` — apiserver pagination).
+
+        Returns ``(items, list_rv, next_continue)``; an empty
+        ``next_continue`` means the list is complete.  The continue token
+        encodes the last returned (namespace, name): like a real apiserver
+        the second chunk continues *after* that key in (ns, name) order.
+        (A real apiserver additionally serves continues from an etcd
+        snapshot; here later chunks see current state, which client-go
+        tolerates — documented in docs/testing.md.)
+        """
+        import base64
+
+        with self._lock:
+            items = self.list(
+                api_version, kind, namespace=namespace,
+                label_selector=label_selector, field_selector=field_selector,
+            )
+            rv = str(self._rv_counter)
+            if continue_token:
+                try:
+                    last = json.loads(
+                        base64.urlsafe_b64decode(continue_token.encode()).decode()
+                    )
+                    last_key = (last["ns"], last["name"])
+                except Exception:
+                    raise BadRequestError(
+                        "invalid continue token"
+                    ) from None
+                items = [
+                    o for o in items
+                    if (meta.namespace(o), meta.name(o)) > last_key
+                ]
+            next_token = ""
+            if limit and len(items) > limit:
+                items = items[:limit]
+                tail = items[-1]
+                next_token = base64.urlsafe_b64encode(json.dumps({
+                    "ns": meta.namespace(tail), "name": meta.name(tail),
+                }).encode()).decode()
+            return items, rv, next_token
 
     def update(self, obj: K8sObject) -> K8sObject:
         obj = meta.deep_copy(obj)

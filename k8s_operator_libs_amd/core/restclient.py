@@ -226,19 +226,35 @@ class RestClient(Client):
             label_selector=label_selector, field_selector=field_selector,
         )[0]
 
+    #: LIST chunk size (client-go reflector uses 500; 0 disables pagination)
+    LIST_PAGE_SIZE = 500
+
     def list_with_meta(self, api_version, kind, namespace=None,
                        label_selector="", field_selector=""):
+        """Chunked LIST following ``continue`` tokens (the client-go
+        pager), returning all items plus the first chunk's list RV."""
         params = {}
         if label_selector:
             params["labelSelector"] = label_selector
         if field_selector:
             params["fieldSelector"] = field_selector
+        if self.LIST_PAGE_SIZE:
+            params["limit"] = str(self.LIST_PAGE_SIZE)
         path = self._collection_path(api_version, kind, namespace or "")
-        resp = self._request("GET", path, params=params)
-        self._raise_for(resp)
-        body = resp.json()
-        rv = (body.get("metadata") or {}).get("resourceVersion")
-        return body.get("items", []), rv
+        items: list = []
+        rv = None
+        while True:
+            resp = self._request("GET", path, params=params)
+            self._raise_for(resp)
+            body = resp.json()
+            md = body.get("metadata") or {}
+            if rv is None:
+                rv = md.get("resourceVersion")
+            items.extend(body.get("items", []))
+            cont = md.get("continue")
+            if not cont:
+                return items, rv
+            params["continue"] = cont
 
     def create(self, obj):
         api_version, kind = meta.api_version(obj), meta.kind(obj)
@@ -263,8 +279,16 @@ class RestClient(Client):
         self._raise_for(resp)
         return resp.json()
 
-    def delete(self, api_version, kind, name, namespace=""):
-        resp = self._request("DELETE", self._object_path(api_version, kind, name, namespace))
+    def delete(self, api_version, kind, name, namespace="", grace_period_seconds=None):
+        kw = {}
+        if grace_period_seconds is not None:
+            kw["content"] = json.dumps({
+                "apiVersion": "v1", "kind": "DeleteOptions",
+                "gracePeriodSeconds": int(grace_period_seconds),
+            })
+        resp = self._request(
+            "DELETE", self._object_path(api_version, kind, name, namespace), **kw
+        )
         self._raise_for(resp)
 
     def patch_status(self, api_version, kind, name, status, namespace=""):

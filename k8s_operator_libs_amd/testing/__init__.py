@@ -130,7 +130,10 @@ class DaemonSetBuilder:
             "kind": "DaemonSet",
             "metadata": {"name": name, "namespace": namespace, "labels": labels},
             "spec": {"selector": {"matchLabels": labels},
-                     "template": {"metadata": {"labels": labels}}},
+                     "template": {"metadata": {"labels": labels},
+                                  "spec": {"containers": [
+                                      {"name": "driver",
+                                       "image": "amdgpu-dkms:latest"}]}}},
             "status": {"desiredNumberScheduled": 0, "numberMisscheduled": 0},
         }
 
@@ -329,3 +332,158 @@ class SimMaintenanceOperator:
                         ]}},
                         meta.namespace(live),
                     )
+
+class ClientHookAdapter:
+    """Give any :class:`~k8s_operator_libs_amd.core.client.Client` the
+    FakeCluster surface the simulated actors use (CRUD passthrough plus
+    ``add_change_hook``), feeding hooks from watch streams instead of
+    in-process mutation hooks.
+
+    This is what lets the same :class:`SimDaemonSetController` /
+    :class:`SimMaintenanceOperator` / :class:`SimKubelet` drive a REAL
+    kube-apiserver (tests/test_conformance.py) where no change-hook exists.
+    Two real-apiserver differences are papered over:
+
+    - ``create`` with an inline ``status`` (the builder convention, matching
+      envtest's forced Status().Update pattern, reference
+      upgrade_suit_test.go:344-355) becomes create + status patch, because a
+      real apiserver drops status on create;
+    - events arrive asynchronously, so hooks must be level-triggered (all of
+      the in-repo sims are).
+    """
+
+    def __init__(self, client, kinds=(("v1", "Pod"),)):
+        self._client = client
+        self._hooks = []
+        self._lock = threading.RLock()
+        self._stopped = threading.Event()
+        self._watch_threads = []
+        self._watches = []
+        for api_version, kind in kinds:
+            w = client.watch(api_version, kind)
+            self._watches.append(w)
+            t = threading.Thread(
+                target=self._pump, args=(w, api_version, kind), daemon=True
+            )
+            t.start()
+            self._watch_threads.append(t)
+
+    def _pump(self, watch, api_version, kind):
+        while not self._stopped.is_set():
+            item = watch.next(timeout=0.2)
+            if item is None:
+                alive = getattr(watch, "alive", None)
+                if alive is not None and not alive():
+                    if self._stopped.is_set():
+                        return
+                    try:
+                        watch = self._client.watch(api_version, kind)
+                    except Exception:
+                        self._stopped.wait(0.5)
+                continue
+            event_type, obj = item
+            if event_type in ("BOOKMARK", "ERROR") or obj is None:
+                continue
+            with self._lock:
+                hooks = list(self._hooks)
+            for hook in hooks:
+                try:
+                    hook(event_type, obj)
+                except Exception:
+                    import logging
+
+                    logging.getLogger(__name__).exception("sim hook failed")
+
+    def add_change_hook(self, hook):
+        with self._lock:
+            self._hooks.append(hook)
+
+    def stop(self):
+        self._stopped.set()
+        for w in self._watches:
+            w.stop()
+
+    # -- FakeCluster-flavoured CRUD over the client ---------------------------
+
+    def create(self, obj):
+        status = obj.pop("status", None)
+        created = self._client.create(obj)
+        if status:
+            created = self._client.patch_status(
+                meta.api_version(created), meta.kind(created),
+                meta.name(created), status, meta.namespace(created),
+            )
+        return created
+
+    def get(self, api_version, kind, name, namespace=""):
+        return self._client.get(api_version, kind, name, namespace)
+
+    def list(self, api_version, kind, namespace=None, label_selector="",
+             field_selector=""):
+        return self._client.list(api_version, kind, namespace=namespace,
+                                 label_selector=label_selector,
+                                 field_selector=field_selector)
+
+    def patch(self, api_version, kind, name, patch, namespace=""):
+        if set(patch) == {"status"}:
+            # sims patch status through the subresource on real servers
+            return self._client.patch_status(api_version, kind, name,
+                                             patch["status"], namespace)
+        return self._client.patch(api_version, kind, name, patch, namespace)
+
+    def delete(self, api_version, kind, name, namespace=""):
+        self._client.delete(api_version, kind, name, namespace)
+
+    def evict_pod(self, name, namespace):
+        self._client.evict_pod(name, namespace)
+
+
+class SimKubelet:
+    """The kubelet role a control-plane-only cluster is missing.
+
+    envtest boots no kubelet, so (a) a deleted pod lingers Terminating
+    forever (graceful deletion waits for kubelet confirmation) and (b) fresh
+    pods never go Running/Ready.  This sim completes both halves: it
+    force-deletes pods carrying a deletionTimestamp (the kubelet's grace-0
+    confirmation) and patches phase=Running + ready containerStatuses onto
+    new pods.  On FakeCluster it is unnecessary (deletes are immediate,
+    builders set status) but harmless."""
+
+    def __init__(self, cluster, ready=True):
+        self.cluster = cluster
+        self.ready = ready
+        self._lock = threading.RLock()
+        cluster.add_change_hook(self._on_change)
+
+    def _on_change(self, event_type, obj):
+        if meta.kind(obj) != "Pod" or event_type == "DELETED":
+            return
+        with self._lock:
+            md = obj.get("metadata", {})
+            name, ns = md.get("name", ""), md.get("namespace", "default")
+            if "deletionTimestamp" in md:
+                client = getattr(self.cluster, "_client", None)
+                try:
+                    if client is not None:
+                        client.delete("v1", "Pod", name, ns,
+                                      grace_period_seconds=0)
+                    else:
+                        self.cluster.delete("v1", "Pod", name, ns)
+                except Exception:
+                    pass
+                return
+            status = obj.get("status", {})
+            if status.get("phase") != "Running" or not status.get("containerStatuses"):
+                containers = obj.get("spec", {}).get("containers", [])
+                try:
+                    self.cluster.patch("v1", "Pod", name, {"status": {
+                        "phase": "Running",
+                        "containerStatuses": [
+                            {"name": c.get("name", "c"), "ready": self.ready,
+                             "restartCount": 0,
+                             "state": {"running": {}}}
+                            for c in containers
+                        ],
+                    }}, ns)
+                except Exception:
+                    pass
