@@ -3,9 +3,15 @@
 
 Measures the metric BASELINE.json names — rolling-upgrade wall-clock and
 reconcile p50 for an 8-node amdgpu driver bump with maxParallelUpgrades=2,
-cordon+drain of synthetic ``amd.com/gpu`` pods — on a synthetic cluster
-served by the in-memory apiserver (this library's envtest).  The reference
-publishes no numbers, so these runs establish the baseline.
+cordon+drain of synthetic ``amd.com/gpu`` pods.  The reference publishes no
+numbers, so these runs establish the baseline.
+
+The HEADLINE substrate is the production-shaped stack: state machine ->
+CachedClient (watch-fed informers) -> RestClient -> HTTP apiserver
+(``--substrate cached``, the default).  ``--substrate http`` drops the
+informer cache (direct REST), ``--substrate inproc`` is the round-1
+in-process dict store, kept for continuity; a single run reports all three
+(secondary substrates at reduced step count).
 
 One *step* = one complete rolling upgrade of an 8-node cluster: every node
 travels upgrade-required -> cordon -> wait-for-jobs -> pod-deletion -> drain
@@ -22,10 +28,12 @@ Scaling is weak: each rank runs an independent 8-node cluster upgrade
 Flags beyond the driver contract (--gpus/--steps/--warmup):
   --nodes N        simulated cluster size per rank (default 8)
   --max-parallel   rolling window (default 2, the BASELINE config)
+  --substrate      cached (default, headline) | http | inproc
   --mode           inplace (default) | requestor (maintenance-operator
-                   delegation with a simulated maintenance operator)
+                   delegation) | anic (BASELINE config #4: NIC/xGMI driver,
+                   wait-for-jobs + xGMI-fabric validation hooks)
   --no-converge    strict reference semantics (one transition per tick)
-  --no-gpu-validate  skip the native MFMA check in the validation step
+  --no-gpu-validate  skip the native MFMA/xGMI checks in the validation step
 """
 
 from __future__ import annotations
@@ -40,38 +48,64 @@ from k8s_operator_libs_amd.api.upgrade.v1alpha1 import DriverUpgradePolicySpec
 from k8s_operator_libs_amd.core import FakeClient
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
-from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
+from k8s_operator_libs_amd.upgrade.state_manager import (
+    BuildStateError,
+    ClusterUpgradeStateManager,
+)
 
 DRIVER_NS = "amd-gpu-operator"
-DRIVER_LABELS = {"app": "amdgpu-driver-daemonset"}
 VALIDATOR_SELECTOR = "app=amd-gpu-validator"
+COMM_JOB_SELECTOR = "app=comm-job"
+
+
+def _driver_labels(mode: str) -> dict:
+    name = "anic" if mode == "anic" else "amdgpu"
+    return {"app": f"{name}-driver-daemonset"}
+
+
+def _driver_ds_name(mode: str) -> str:
+    return "anic-driver" if mode == "anic" else "amdgpu-driver"
 
 
 # ---------------------------------------------------------------------------
 # synthetic cluster construction
 # ---------------------------------------------------------------------------
 
-def _make_cluster(client, n_nodes, old_hash, new_hash, gpu_pods_per_node=2):
-    c = client.cluster
-    labels = dict(DRIVER_LABELS)
-    ds = c.create({
+def _make_cluster(sim, n_nodes, old_hash, new_hash, gpu_pods_per_node=2,
+                  mode="inplace"):
+    """Create the synthetic driver cluster directly in the backing store
+    (cluster-side setup, the envtest Status().Update pattern); the library
+    itself talks to it through whichever substrate is under test.  Returns
+    (ds, cleanup) where cleanup removes every object this call created."""
+    labels = _driver_labels(mode)
+    ds_name = _driver_ds_name(mode)
+    created = []  # (api_version, kind, name, namespace)
+
+    def mk(obj):
+        out = sim.create(obj)
+        created.append((obj["apiVersion"], obj["kind"],
+                        out["metadata"]["name"],
+                        out["metadata"].get("namespace", "")))
+        return out
+
+    ds = mk({
         "apiVersion": "apps/v1", "kind": "DaemonSet",
-        "metadata": {"name": "amdgpu-driver", "namespace": DRIVER_NS,
+        "metadata": {"name": ds_name, "namespace": DRIVER_NS,
                      "labels": labels},
         "spec": {"selector": {"matchLabels": labels},
                  "template": {"metadata": {"labels": labels}}},
         "status": {"desiredNumberScheduled": n_nodes},
     })
     for rev, h in ((1, old_hash), (2, new_hash)):
-        c.create({
+        mk({
             "apiVersion": "apps/v1", "kind": "ControllerRevision",
-            "metadata": {"name": f"amdgpu-driver-{h}", "namespace": DRIVER_NS,
+            "metadata": {"name": f"{ds_name}-{h}", "namespace": DRIVER_NS,
                          "labels": dict(labels)},
             "revision": rev,
         })
     for i in range(n_nodes):
         node = f"mi355x-{i}"
-        c.create({
+        mk({
             "apiVersion": "v1", "kind": "Node",
             "metadata": {"name": node, "labels": {}, "annotations": {}},
             "spec": {},
@@ -80,24 +114,24 @@ def _make_cluster(client, n_nodes, old_hash, new_hash, gpu_pods_per_node=2):
         })
         pod_labels = dict(labels)
         pod_labels["controller-revision-hash"] = old_hash
-        c.create({
+        mk({
             "apiVersion": "v1", "kind": "Pod",
-            "metadata": {"name": f"amdgpu-driver-{node}", "namespace": DRIVER_NS,
+            "metadata": {"name": f"{ds_name}-{node}", "namespace": DRIVER_NS,
                          "labels": pod_labels,
                          "ownerReferences": [{"apiVersion": "apps/v1",
                                               "kind": "DaemonSet",
-                                              "name": "amdgpu-driver",
+                                              "name": ds_name,
                                               "uid": ds["metadata"]["uid"],
                                               "controller": True}]},
             "spec": {"nodeName": node,
-                     "containers": [{"name": "driver", "image": "amdgpu-dkms:old"}]},
+                     "containers": [{"name": "driver", "image": f"{ds_name}:old"}]},
             "status": {"phase": "Running",
                        "containerStatuses": [{"name": "driver", "ready": True,
                                               "restartCount": 0}]},
         })
         # synthetic GPU workload pods that must be evicted before the bump
         for j in range(gpu_pods_per_node):
-            c.create({
+            mk({
                 "apiVersion": "v1", "kind": "Pod",
                 "metadata": {"name": f"train-{node}-{j}", "namespace": "default",
                              "labels": {"app": "training"},
@@ -113,8 +147,22 @@ def _make_cluster(client, n_nodes, old_hash, new_hash, gpu_pods_per_node=2):
                            "containerStatuses": [{"name": "t", "ready": True,
                                                   "restartCount": 0}]},
             })
+        if mode == "anic":
+            # in-flight collective-communication jobs the NIC upgrade must
+            # wait out (BASELINE config #4's wait-for-jobs hook)
+            mk({
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": f"comm-{node}", "namespace": "default",
+                             "labels": {"app": "comm-job"}},
+                "spec": {"nodeName": node,
+                         "containers": [{"name": "allreduce", "image": "rccl-job"}]},
+                "status": {"phase": "Running",
+                           "containerStatuses": [{"name": "allreduce",
+                                                  "ready": True,
+                                                  "restartCount": 0}]},
+            })
         # validation pod (not ready until the GPU health check passes)
-        c.create({
+        mk({
             "apiVersion": "v1", "kind": "Pod",
             "metadata": {"name": f"validator-{node}", "namespace": DRIVER_NS,
                          "labels": {"app": "amd-gpu-validator"}},
@@ -124,28 +172,45 @@ def _make_cluster(client, n_nodes, old_hash, new_hash, gpu_pods_per_node=2):
                        "containerStatuses": [{"name": "v", "ready": False,
                                               "restartCount": 0}]},
         })
-    return ds
+
+    def cleanup():
+        # driver pods may have been recreated under the same name; workload
+        # pods may already be evicted — delete whatever still exists
+        for api_version, kind, name, ns in reversed(created):
+            try:
+                sim.delete(api_version, kind, name, ns)
+            except Exception:
+                pass
+
+    return ds, cleanup
 
 
 class _DsController:
     """Recreates deleted driver pods with the new revision (the DaemonSet
-    controller + kubelet role, like envtest tests do by hand)."""
+    controller + kubelet role, like envtest tests do by hand).  Installed
+    once per substrate; ``retarget`` points it at each iteration's
+    DaemonSet."""
 
-    def __init__(self, cluster, ds, new_hash):
+    def __init__(self, cluster, ds, new_hash, labels):
         import threading
 
         self.cluster, self.ds, self.new_hash = cluster, ds, new_hash
+        self.labels = labels
         self._lock = threading.Lock()
         cluster.add_change_hook(self._on_change)
+
+    def retarget(self, ds):
+        with self._lock:
+            self.ds = ds
 
     def _on_change(self, event_type, obj):
         if event_type != "DELETED" or obj.get("kind") != "Pod":
             return
-        refs = obj.get("metadata", {}).get("ownerReferences") or []
-        if not refs or refs[0].get("uid") != self.ds["metadata"]["uid"]:
-            return
         with self._lock:
-            labels = dict(DRIVER_LABELS)
+            refs = obj.get("metadata", {}).get("ownerReferences") or []
+            if not refs or refs[0].get("uid") != self.ds["metadata"]["uid"]:
+                return
+            labels = dict(self.labels)
             labels["controller-revision-hash"] = self.new_hash
             self.cluster.create({
                 "apiVersion": "v1", "kind": "Pod",
@@ -154,7 +219,7 @@ class _DsController:
                              "labels": labels,
                              "ownerReferences": refs},
                 "spec": {"nodeName": obj["spec"].get("nodeName", ""),
-                         "containers": [{"name": "driver", "image": "amdgpu-dkms:new"}]},
+                         "containers": [{"name": "driver", "image": "driver:new"}]},
                 "status": {"phase": "Running",
                            "containerStatuses": [{"name": "driver", "ready": True,
                                                   "restartCount": 0}]},
@@ -224,24 +289,73 @@ class _MaintenanceOperatorSim:
                                    live["metadata"].get("namespace", ""))
 
 
-def _run_gpu_validation(device):
-    """The validation-pod payload: native gfx950 MFMA smoke on this rank's
-    GPU.  Raises if the native extension is missing on a GPU box."""
+def _run_gpu_validation(device, mode="inplace"):
+    """The validation-pod payload on this rank's GPU.  Raises if the native
+    extension is missing on a GPU box.
+
+    - amdgpu modes: native gfx950 MFMA matrix-core smoke.
+    - anic mode (NIC/xGMI driver, BASELINE config #4): probe the xGMI
+      fabric — peer reachability + per-link p2p bandwidth
+      (native/gpu_validator.hip xgmi_p2p_probe); on a single-GPU box there
+      are no peers, so fall back to the MFMA smoke (still real GPU work)
+      and note it.
+    """
     from k8s_operator_libs_amd.validation import load_native_validator
 
     native = load_native_validator()
     if native is None:
         raise RuntimeError("GPU validation requested but native validator missing")
+    if mode == "anic":
+        links = native.xgmi_p2p_probe(device, 64.0, 2)
+        if links:
+            bad = [l for l in links if not l.get("accessible") or "error" in l]
+            if bad:
+                raise RuntimeError(f"xGMI validation failed: {bad}")
+            return min(l.get("bandwidth_gbps", 0.0) for l in links)
     err = native.mfma_f32_check(device)
     if err > 1e-6:
         raise RuntimeError(f"MFMA validation failed: err={err}")
     return err
 
 
+def _build_substrate(substrate):
+    """Returns (client, sim_cluster, close_fn) for the requested substrate.
+
+    - ``inproc``: direct dict-backed FakeClient (round-1 continuity number)
+    - ``http``:   RestClient -> HTTP apiserver (wire-level, no cache)
+    - ``cached``: CachedClient informers -> RestClient -> HTTP apiserver
+                  (the production architecture; HEADLINE)
+    """
+    if substrate == "inproc":
+        client = FakeClient()
+        return client, client.cluster, lambda: None
+    from k8s_operator_libs_amd.core.apiserver import start_apiserver
+    from k8s_operator_libs_amd.core.restclient import RestClient
+
+    handle = start_apiserver()
+    rest = RestClient(handle.url)
+    if substrate == "http":
+        def close():
+            rest.close()
+            handle.stop()
+        return rest, handle.cluster, close
+    if substrate == "cached":
+        from k8s_operator_libs_amd.core.cache import CachedClient
+
+        cached = CachedClient(rest)
+
+        def close():
+            cached.stop()
+            rest.close()
+            handle.stop()
+        return cached, handle.cluster, close
+    raise ValueError(f"unknown substrate {substrate!r}")
+
+
 def run_rolling_upgrade_benchmark(
     n_nodes=8, steps=10, warmup=2, max_parallel=2, gpu_validate=False,
     device=0, gpu_pods_per_node=2, converge=True,
-    mode="inplace",
+    mode="inplace", substrate="cached",
 ):
     """Run `warmup` untimed + `steps` timed full rolling upgrades; returns a
     result dict (single-process path; bench main() adds distribution).
@@ -254,74 +368,151 @@ def run_rolling_upgrade_benchmark(
     from k8s_operator_libs_amd.upgrade.requestor import RequestorOptions
     from k8s_operator_libs_amd.upgrade.state_manager import StateOptions
 
-    policy = DriverUpgradePolicySpec.model_validate({
+    policy_doc = {
         "autoUpgrade": True,
         "maxParallelUpgrades": max_parallel,
         "maxUnavailable": "50%",
         "podDeletion": {"force": False, "deleteEmptyDir": True},
         "drain": {"enable": True, "timeoutSeconds": 300},
-    })
+    }
+    if mode == "anic":
+        # BASELINE config #4: NIC upgrades wait out in-flight communication
+        # jobs before quiescing the node (docs/automatic-ofed-upgrade.md)
+        policy_doc["waitForCompletion"] = {
+            "podSelector": COMM_JOB_SELECTOR, "timeoutSeconds": 300,
+        }
+    policy = DriverUpgradePolicySpec.model_validate(policy_doc)
+
+    prev_driver = util.get_driver_name()
+    util.set_driver_name("anic" if mode == "anic" else "amdgpu")
     state_key = util.get_upgrade_state_label_key()
+    driver_labels = _driver_labels(mode)
 
     wall_times = []
     tick_times = []
     ticks_per_upgrade = []
     completed = 0
 
-    for it in range(warmup + steps):
-        timed = it >= warmup
-        client = FakeClient()
-        ds = _make_cluster(client, n_nodes, "oldrev", "newrev", gpu_pods_per_node)
-        _DsController(client.cluster, ds, "newrev")
-        registry = MetricsRegistry()
-        options = None
-        if mode == "requestor":
-            _MaintenanceOperatorSim(client.cluster)
-            options = StateOptions(requestor=RequestorOptions(
-                use_maintenance_operator=True,
-                requestor_id="amd.gpu.operator",
-                namespace="default",
-            ))
-        manager = (
-            ClusterUpgradeStateManager(client, metrics=registry, options=options)
-            .with_pod_deletion_enabled(gpu_pod_deletion_filter)
-            .with_validation_enabled(VALIDATOR_SELECTOR)
-        )
-        t0 = time.perf_counter()
-        rounds = 0
-        while True:
-            state = manager.reconcile(DRIVER_NS, DRIVER_LABELS, policy,
-                                      converge=converge)
-            rounds += 1
-            # validation pods: run the real GPU check, then mark Ready
-            for ns_ in state.nodes_in(consts.UPGRADE_STATE_VALIDATION_REQUIRED):
-                node_name = ns_.node["metadata"]["name"]
-                if gpu_validate:
-                    _run_gpu_validation(device)
-                client.cluster.patch(
-                    "v1", "Pod", f"validator-{node_name}",
-                    {"status": {"containerStatuses": [
-                        {"name": "v", "ready": True, "restartCount": 0}]}},
-                    DRIVER_NS,
-                )
-            done = sum(
-                1 for n in client.list_nodes()
-                if n["metadata"]["labels"].get(state_key) == consts.UPGRADE_STATE_DONE
+    client, sim, close_substrate = _build_substrate(substrate)
+    ctrl = _DsController(sim, {"metadata": {"uid": "-"}}, "newrev", driver_labels)
+    if mode == "requestor":
+        _MaintenanceOperatorSim(sim)
+
+    try:
+        for it in range(warmup + steps):
+            timed = it >= warmup
+            ds, cleanup = _make_cluster(sim, n_nodes, "oldrev", "newrev",
+                                        gpu_pods_per_node, mode=mode)
+            ctrl.retarget(ds)
+            # eventually-consistent substrates: wait until the client view
+            # reflects the fully-built cluster before the timed region (the
+            # production analogue is controller-runtime WaitForCacheSync)
+            deadline = time.monotonic() + 30
+            while time.monotonic() < deadline:
+                try:
+                    dss = client.list_daemonsets(namespace=DRIVER_NS,
+                                                 label_selector=",".join(
+                                                     f"{k}={v}" for k, v in
+                                                     driver_labels.items()))
+                    pods = client.list_pods(namespace=DRIVER_NS)
+                    if dss and len([p for p in pods
+                                    if p["metadata"]["labels"].get(
+                                        "controller-revision-hash")]) >= n_nodes \
+                            and len(client.list_nodes()) >= n_nodes:
+                        break
+                except Exception:
+                    pass
+                time.sleep(0.001)
+            registry = MetricsRegistry()
+            options = None
+            if mode == "requestor":
+                options = StateOptions(requestor=RequestorOptions(
+                    use_maintenance_operator=True,
+                    requestor_id="amd.gpu.operator",
+                    namespace="default",
+                ))
+            manager = (
+                ClusterUpgradeStateManager(client, metrics=registry,
+                                           options=options)
+                .with_pod_deletion_enabled(gpu_pod_deletion_filter)
+                .with_validation_enabled(VALIDATOR_SELECTOR)
             )
-            if done == n_nodes:
-                break
-            if rounds > 60 * n_nodes:
-                raise RuntimeError(f"upgrade did not converge after {rounds} rounds")
-        elapsed = time.perf_counter() - t0
-        ticks = registry.reconcile_duration.count
-        if timed:
-            tick_times.extend(registry.reconcile_duration.samples())
-        if timed:
-            wall_times.append(elapsed)
-            ticks_per_upgrade.append(ticks)
-            completed += 1
+            t0 = time.perf_counter()
+            rounds = 0
+            last_done = -1
+            while True:
+                try:
+                    state = manager.reconcile(DRIVER_NS, driver_labels, policy,
+                                              converge=converge)
+                except BuildStateError:
+                    # transient cache view of a driver pod mid-recreation;
+                    # the reference requeues the reconcile on error
+                    # (upgrade_state.go:128-131) — retry next round
+                    rounds += 1
+                    time.sleep(0.0005)
+                    continue
+                rounds += 1
+                if mode == "anic":
+                    # the in-flight communication jobs complete while the
+                    # node waits (wait-for-jobs hook observes Succeeded)
+                    for ns_ in state.nodes_in(
+                            consts.UPGRADE_STATE_WAIT_FOR_JOBS_REQUIRED):
+                        node_name = ns_.node["metadata"]["name"]
+                        try:
+                            sim.patch(
+                                "v1", "Pod", f"comm-{node_name}",
+                                {"status": {"phase": "Succeeded",
+                                            "containerStatuses": [
+                                                {"name": "allreduce",
+                                                 "ready": False,
+                                                 "restartCount": 0}]}},
+                                "default",
+                            )
+                        except Exception:
+                            pass
+                # validation pods: run the real GPU check, then mark Ready
+                for ns_ in state.nodes_in(consts.UPGRADE_STATE_VALIDATION_REQUIRED):
+                    node_name = ns_.node["metadata"]["name"]
+                    if gpu_validate:
+                        _run_gpu_validation(device, mode=mode)
+                    sim.patch(
+                        "v1", "Pod", f"validator-{node_name}",
+                        {"status": {"containerStatuses": [
+                            {"name": "v", "ready": True, "restartCount": 0}]}},
+                        DRIVER_NS,
+                    )
+                done = sum(
+                    1 for n in client.list_nodes()
+                    if n["metadata"]["labels"].get(state_key)
+                    == consts.UPGRADE_STATE_DONE
+                )
+                if done == n_nodes:
+                    break
+                if rounds > 200 * n_nodes:
+                    raise RuntimeError(
+                        f"upgrade did not converge after {rounds} rounds"
+                    )
+                if done == last_done and substrate != "inproc":
+                    # waiting on async workers / informer propagation —
+                    # yield instead of hammering the apiserver
+                    time.sleep(0.0005)
+                last_done = done
+            elapsed = time.perf_counter() - t0
+            ticks = registry.reconcile_duration.count
+            if timed:
+                tick_times.extend(registry.reconcile_duration.samples())
+                wall_times.append(elapsed)
+                ticks_per_upgrade.append(ticks)
+                completed += 1
+            manager.wait_idle()
+            ctrl.retarget({"metadata": {"uid": "-"}})
+            cleanup()
+    finally:
+        util.set_driver_name(prev_driver)
+        close_substrate()
 
     return {
+        "substrate": substrate,
         "upgrades_completed": completed,
         "mean_wall_s": statistics.mean(wall_times) if wall_times else 0.0,
         "wall_times": wall_times,
@@ -351,8 +542,18 @@ def main():
     parser.add_argument("--no-gpu-validate", action="store_true")
     parser.add_argument("--no-converge", action="store_true",
                         help="reference semantics: one state transition per reconcile tick")
-    parser.add_argument("--mode", choices=["inplace", "requestor"], default="inplace",
-                        help="in-place node ops vs maintenance-operator delegation")
+    parser.add_argument("--mode", choices=["inplace", "requestor", "anic"],
+                        default="inplace",
+                        help="in-place node ops | maintenance-operator "
+                             "delegation | NIC/xGMI driver path (config #4)")
+    parser.add_argument("--substrate", choices=["cached", "http", "inproc"],
+                        default="cached",
+                        help="cached = informers->REST->HTTP apiserver "
+                             "(production-shaped, HEADLINE); http = direct "
+                             "REST; inproc = round-1 dict store")
+    parser.add_argument("--no-secondary-substrates", action="store_true",
+                        help="skip the reduced-step runs of the other two "
+                             "substrates (reported for comparison)")
     args = parser.parse_args()
 
     import torch
@@ -385,7 +586,7 @@ def main():
         n_nodes=args.nodes, steps=0, warmup=args.warmup,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
         device=local_rank if have_cuda else 0, converge=converge,
-        mode=args.mode,
+        mode=args.mode, substrate=args.substrate,
     )
     barrier_sync()
     t0 = time.perf_counter()
@@ -393,10 +594,32 @@ def main():
         n_nodes=args.nodes, steps=args.steps, warmup=0,
         max_parallel=args.max_parallel, gpu_validate=gpu_validate,
         device=local_rank if have_cuda else 0, converge=converge,
-        mode=args.mode,
+        mode=args.mode, substrate=args.substrate,
     )
     barrier_sync()
     elapsed = time.perf_counter() - t0
+
+    # comparison runs of the non-headline substrates (reduced step count,
+    # outside the timed region; rank 0 only)
+    substrate_results = {args.substrate: {
+        "mean_wall_s": result["mean_wall_s"],
+        "reconcile_p50_ms": result["reconcile_p50_ms"],
+        "steps": args.steps,
+    }}
+    if rank == 0 and not args.no_secondary_substrates:
+        for other in ("cached", "http", "inproc"):
+            if other == args.substrate:
+                continue
+            r = run_rolling_upgrade_benchmark(
+                n_nodes=args.nodes, steps=min(3, args.steps), warmup=1,
+                max_parallel=args.max_parallel, gpu_validate=False,
+                converge=converge, mode=args.mode, substrate=other,
+            )
+            substrate_results[other] = {
+                "mean_wall_s": r["mean_wall_s"],
+                "reconcile_p50_ms": r["reconcile_p50_ms"],
+                "steps": min(3, args.steps),
+            }
 
     # MAX over ranks for time-like metrics
     if dist is not None:
@@ -414,6 +637,8 @@ def main():
             "metric": "8-node rolling-upgrade wall-clock",
             "value": mean_wall_s,
             "unit": "s",
+            "substrate": args.substrate,
+            "substrates": substrate_results,
             "n_gpus": args.gpus,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -422,10 +647,17 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "fp32",
-            "data": "synthetic (in-memory apiserver, synthetic amd.com/gpu pods; "
-                    "reference publishes no baseline numbers)",
+            "data": "synthetic (synthetic amd.com/gpu pods; substrate="
+                    f"{args.substrate}: "
+                    + ("informer cache -> REST -> HTTP apiserver"
+                       if args.substrate == "cached" else
+                       "REST -> HTTP apiserver" if args.substrate == "http"
+                       else "in-process store")
+                    + "; reference publishes no baseline numbers)",
             "config": {
-                "model": "amdgpu-driver-rolling-upgrade",
+                "model": ("anic" if args.mode == "anic" else "amdgpu")
+                + "-driver-rolling-upgrade",
+                "substrate": args.substrate,
                 "nodes_per_cluster": args.nodes,
                 "maxParallelUpgrades": args.max_parallel,
                 "maxUnavailable": "50%",

@@ -103,6 +103,7 @@ class _Informer:
                 if alive is not None and not alive():
                     # stream dropped: re-watch from the last processed RV —
                     # only a 410 (resume window expired) forces a full relist
+                    self._watch.stop()
                     try:
                         self._watch = self._open_watch()
                         if self._last_rv is None:

@@ -57,3 +57,60 @@ def test_performance_regression_guard():
     )
     assert result["mean_wall_s"] < 2.0, result
     assert result["reconcile_p50_ms"] < 200.0, result
+
+
+def test_bench_substrates_all_converge():
+    """All three substrates complete an upgrade and
+    report the substrate they measured."""
+    import bench
+
+    for substrate in ("inproc", "http", "cached"):
+        result = bench.run_rolling_upgrade_benchmark(
+            n_nodes=2, steps=1, warmup=0, max_parallel=1, gpu_validate=False,
+            substrate=substrate,
+        )
+        assert result["upgrades_completed"] == 1, substrate
+        assert result["substrate"] == substrate
+
+
+def test_bench_anic_mode():
+    """BASELINE config #4: NIC/xGMI driver path with wait-for-jobs hooks."""
+    import bench
+    from k8s_operator_libs_amd.upgrade import util
+
+    result = bench.run_rolling_upgrade_benchmark(
+        n_nodes=2, steps=1, warmup=0, max_parallel=2, gpu_validate=False,
+        mode="anic", substrate="inproc",
+    )
+    assert result["upgrades_completed"] == 1
+    # driver-name global restored after the run
+    assert util.get_driver_name() == "amdgpu"
+
+
+def test_bench_multirank_gloo_world4():
+    """the distributed bench path executes under
+    WORLD_SIZE=4 with the gloo backend — one simulated cluster per rank,
+    MAX-reduce across ranks, exactly one contract JSON line from rank 0 —
+    so the first 8-GPU lease runs the already-proven path."""
+    repo = os.path.join(os.path.dirname(__file__), "..")
+    env = dict(os.environ)
+    env.setdefault("MASTER_ADDR", "127.0.0.1")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run",
+         "--nnodes=1", "--nproc-per-node", "4",
+         "--master-addr", "127.0.0.1", "--master-port", "29517",
+         "bench.py", "--gpus", "4", "--steps", "1", "--warmup", "0",
+         "--nodes", "2", "--substrate", "inproc",
+         "--no-secondary-substrates"],
+        capture_output=True, text=True, cwd=repo, timeout=600, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-3000:]
+    json_lines = [l for l in out.stdout.strip().splitlines()
+                  if l.startswith("{")]
+    assert len(json_lines) == 1, f"expected exactly one JSON line: {json_lines}"
+    data = json.loads(json_lines[0])
+    assert data["n_gpus"] == 4
+    assert data["config"]["parallelism"].startswith("dp4")
+    assert data["value"] > 0
+    # MAX-reduce: the reported wall is >= any single rank could have reported
+    assert data["ms_per_step"] > 0
