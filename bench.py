@@ -67,6 +67,10 @@ def _driver_ds_name(mode: str) -> str:
     return "anic-driver" if mode == "anic" else "amdgpu-driver"
 
 
+# default (amdgpu in-place) labels, importable by the example operator
+DRIVER_LABELS = _driver_labels("inplace")
+
+
 # ---------------------------------------------------------------------------
 # synthetic cluster construction
 # ---------------------------------------------------------------------------

@@ -74,11 +74,10 @@ def run_demo(args):
     log.info("mini-apiserver at %s", handle.url)
     client = RestClient(handle.url)
 
-    class W:
-        cluster = handle.cluster
-
-    ds = bench_mod._make_cluster(W, args.demo_nodes, "oldrev", "newrev")
-    bench_mod._DsController(handle.cluster, ds, "newrev")
+    ds, _cleanup = bench_mod._make_cluster(handle.cluster, args.demo_nodes,
+                                           "oldrev", "newrev")
+    bench_mod._DsController(handle.cluster, ds, "newrev",
+                            bench_mod._driver_labels("inplace"))
 
     options = None
     if args.demo_requestor:
