@@ -51,9 +51,13 @@ bench:
 demo:
 	$(PY) examples/amdgpu_upgrade_operator.py --demo
 
-# Syntax/bytecode sanity over the package (stand-in for golangci-lint).
+# Real linting (golangci-lint stand-in): bytecode sanity + the in-repo AST
+# linter (tools/lint.py — unused imports, mutable defaults, bare excepts,
+# placeholder-less f-strings, print-in-library, missing docstrings, ...;
+# ruff/mypy are not installable in the offline image).
 lint:
-	$(PY) -m compileall -q k8s_operator_libs_amd tests examples bench.py __graft_entry__.py
+	$(PY) -m compileall -q k8s_operator_libs_amd tests examples tools bench.py __graft_entry__.py
+	$(PY) tools/lint.py
 
 coverage:
 	$(PY) -m pytest tests/ -q -m "not gpu" --timeout 300 \

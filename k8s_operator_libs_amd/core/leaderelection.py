@@ -156,6 +156,9 @@ class LeaderElector:
         (it should run until it observes lost leadership / stop), renewing in
         a background thread.  Returns when stopped."""
         self._on_stopped_leading = on_stopped_leading
+        # fires at most once even though both the renew loop (leadership
+        # lost mid-work) and run()'s cleanup path reach it (ADVICE r1)
+        self._stopped_fired = threading.Event()
         while not self._stop.is_set():
             if self._try_acquire_or_renew():
                 logger.info("leader election: %s acquired %s/%s",
@@ -169,10 +172,20 @@ class LeaderElector:
                     self._leading.clear()
                     renewer.join(timeout=self.retry_period * 2)
                     self._release()
-                    if on_stopped_leading is not None:
-                        on_stopped_leading()
+                    self._fire_stopped_leading()
                 return
             self._stop.wait(self.retry_period)
+
+    def _fire_stopped_leading(self) -> None:
+        cb = getattr(self, "_on_stopped_leading", None)
+        fired = getattr(self, "_stopped_fired", None)
+        if cb is None or fired is None or fired.is_set():
+            return
+        fired.set()
+        try:
+            cb()
+        except Exception:
+            logger.exception("on_stopped_leading callback failed")
 
     def _renew_loop(self) -> None:
         while self._leading.is_set() and not self._stop.is_set():
@@ -183,11 +196,6 @@ class LeaderElector:
                 # Leadership lost while the leader's work is still running:
                 # fire the callback so the work is STOPPED (without this a
                 # demoted replica would keep reconciling — split-brain).
-                cb = getattr(self, "_on_stopped_leading", None)
-                if cb is not None:
-                    try:
-                        cb()
-                    except Exception:
-                        logger.exception("on_stopped_leading callback failed")
+                self._fire_stopped_leading()
                 return
             self._stop.wait(self.retry_period)

@@ -133,6 +133,12 @@ def _apply_one(client: Client, crd: K8sObject) -> None:
             if attempt == _CONFLICT_RETRIES:
                 raise
             time.sleep(0.01 * (attempt + 1))
+    # every attempt hit the create/AlreadyExists race: surface it rather
+    # than silently skipping the apply
+    raise CrdUtilError(
+        f"CRD {name}: create kept racing with a concurrent applier "
+        f"after {_CONFLICT_RETRIES + 1} attempts"
+    )
 
 
 def delete_crds(client: Client, crds: List[K8sObject]) -> None:

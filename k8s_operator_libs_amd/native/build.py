@@ -39,7 +39,7 @@ def build_jsonops(force: bool = False, verbose: bool = False) -> str:
         JSONOPS_SOURCE, "-o", JSONOPS_OUTPUT,
     ]
     if verbose:
-        print(" ".join(cmd))
+        print(" ".join(cmd))  # noqa: T201 (CLI/build output)
     subprocess.run(cmd, check=True, capture_output=not verbose)
     return JSONOPS_OUTPUT
 
@@ -66,11 +66,11 @@ def build(force: bool = False, verbose: bool = False) -> str:
         OUTPUT,
     ]
     if verbose:
-        print(" ".join(cmd))
+        print(" ".join(cmd))  # noqa: T201 (CLI/build output)
     subprocess.run(cmd, check=True, capture_output=not verbose)
     return OUTPUT
 
 
 if __name__ == "__main__":
     build(force=True, verbose=True)
-    print(f"built {OUTPUT}")
+    print(f"built {OUTPUT}")  # noqa: T201 (CLI/build output)

@@ -138,12 +138,46 @@ class UpgradeController:
         finally:
             self.stop_watches()
 
+    def run_with_leader_election(
+        self,
+        lease_name: str = "amd-gpu-operator-upgrade",
+        lease_namespace: str = "default",
+        identity: str = "",
+        **elector_kwargs,
+    ) -> None:
+        """Run the reconcile loop under Lease-based leader election (the
+        controller-runtime manager's election): only the
+        lease holder reconciles; on lost leadership the loop stops so a
+        demoted replica can never split-brain the state machine.  Blocks
+        until :meth:`stop` or leadership is lost."""
+        from ..core.leaderelection import LeaderElector
+
+        elector = LeaderElector(
+            self.manager.common.client, lease_name,
+            namespace=lease_namespace,
+            **({"identity": identity} if identity else {}),
+            **elector_kwargs,
+        )
+        self._elector = elector
+
+        def stop_all():
+            self.stop()
+
+        try:
+            elector.run(on_started_leading=self.run,
+                        on_stopped_leading=stop_all)
+        finally:
+            elector.stop()
+
     def wake(self) -> None:
         self._wake.set()
 
     def stop(self) -> None:
         self._stop.set()
         self._wake.set()
+        elector = getattr(self, "_elector", None)
+        if elector is not None:
+            elector.stop()
 
     def stop_watches(self) -> None:
         self._stop.set()

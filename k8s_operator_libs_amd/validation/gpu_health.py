@@ -152,7 +152,7 @@ def main() -> int:
     import sys
 
     report = gpu_health_check(require_gpu=True, deep="--deep" in sys.argv)
-    print(json.dumps(report, indent=2, default=str))
+    print(json.dumps(report, indent=2, default=str))  # noqa: T201 (CLI/build output)
     return 0 if report["healthy"] else 1
 
 

@@ -7,7 +7,6 @@ import time
 import pytest
 
 from k8s_operator_libs_amd.core.cache import CachedClient
-from k8s_operator_libs_amd.core.client import FakeClient
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.node_state_provider import NodeUpgradeStateProvider
 from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager

@@ -18,7 +18,6 @@ _SCALE = max(1, int(_os.environ.get("HYPOTHESIS_SCALE", "1")))
 
 from k8s_operator_libs_amd.core import FakeClient
 from k8s_operator_libs_amd.upgrade import consts, util
-from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
 from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
 
 from builders import PodBuilder

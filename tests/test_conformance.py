@@ -36,8 +36,7 @@ from k8s_operator_libs_amd.testing import (
     ClientHookAdapter,
     DaemonSetBuilder,
     NodeBuilder,
-    PodBuilder,
-    SimDaemonSetController,
+        SimDaemonSetController,
     SimKubelet,
     SimMaintenanceOperator,
     driver_pod_for,

@@ -99,3 +99,63 @@ def test_requestor_controller_wakes_on_nodemaintenance_ready(client):
     controller.stop()
     assert done.get("ok") is True
     assert state_of(client, "node-0") == consts.UPGRADE_STATE_DONE
+
+
+class TestControllerLeaderElection:
+    """UpgradeController under Lease-based election."""
+
+    def test_only_leader_reconciles_and_failover(self):
+        import threading
+        import time
+
+        from k8s_operator_libs_amd.core.client import FakeClient
+        from k8s_operator_libs_amd.upgrade.controller import UpgradeController
+        from k8s_operator_libs_amd.upgrade.state_manager import (
+            ClusterUpgradeStateManager,
+        )
+
+        client = FakeClient()
+        setup_cluster(client, n_nodes=1)
+
+        def make(identity):
+            manager = ClusterUpgradeStateManager(FakeClient(client.cluster))
+            return UpgradeController(
+                manager, DRIVER_NS, DRIVER_LABELS,
+                policy(drainSpec={"enable": True}), resync_seconds=0.05,
+            ), identity
+
+        c1, _ = make("replica-1")
+        c2, _ = make("replica-2")
+        threads = []
+        for ctrl, ident in ((c1, "replica-1"), (c2, "replica-2")):
+            t = threading.Thread(
+                target=ctrl.run_with_leader_election,
+                kwargs=dict(lease_name="test-upgrade-lease",
+                            identity=ident,
+                            lease_duration=0.5, retry_period=0.05),
+                daemon=True,
+            )
+            t.start()
+            threads.append(t)
+
+        # exactly one replica wins and reconciles
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if c1.reconcile_count + c2.reconcile_count > 2:
+                break
+            time.sleep(0.05)
+        leader, standby = (c1, c2) if c1.reconcile_count else (c2, c1)
+        assert leader.reconcile_count > 0
+        assert standby.reconcile_count == 0, "standby replica reconciled"
+
+        # leader steps down -> standby takes over
+        leader.stop()
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if standby.reconcile_count > 0:
+                break
+            time.sleep(0.05)
+        assert standby.reconcile_count > 0, "no failover to standby"
+        standby.stop()
+        for t in threads:
+            t.join(timeout=5)

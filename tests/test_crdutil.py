@@ -194,7 +194,6 @@ class TestApplyRaces:
     def test_update_conflict_retries_until_success(self, client, tmp_path):
         """A concurrent writer bumping the CRD between our GET and UPDATE
         forces RetryOnConflict behaviour (crdutil.go:214-249)."""
-        from k8s_operator_libs_amd.crdutil import crdutil as crdmod
 
         p = write(str(tmp_path / "w.yaml"),
                   CRD_TMPL.format(group="amd.com", kind="Widget", plural="widgets"))

@@ -9,14 +9,12 @@ conflict storms, operator restart mid-flight — and assert recovery."""
 
 import threading
 
-import pytest
 
-from k8s_operator_libs_amd.api.upgrade.v1alpha1 import DriverUpgradePolicySpec
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
 from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
 
-from builders import DRIVER_LABELS, DRIVER_NS, NodeBuilder, PodBuilder
+from builders import DRIVER_LABELS, DRIVER_NS, PodBuilder
 from simenv import SimDaemonSetController
 from test_state_manager import policy, setup_cluster, state_of
 

@@ -11,10 +11,18 @@ pytestmark = pytest.mark.gpu
 
 @pytest.fixture(scope="module")
 def native():
+    import os
+
     from k8s_operator_libs_amd.validation import GpuHealthError, load_native_validator
 
     mod = load_native_validator()
     if mod is None:
+        if not os.path.exists("/dev/kfd"):
+            # CPU-only machine running an unfiltered `pytest tests/`: skip.
+            # On a real GPU box (ROCm kfd present) a missing native
+            # extension stays a HARD failure — a silent skip there would
+            # green-wash the exact fallback the driver checks for.
+            pytest.skip("no GPU on this machine (missing /dev/kfd)")
         raise GpuHealthError(
             "native validator must be present on a GPU box - refusing to skip"
         )

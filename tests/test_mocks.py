@@ -4,8 +4,7 @@ managers — the reference's mockery-mock isolation pattern
 
 from k8s_operator_libs_amd.api.upgrade.v1alpha1 import (
     DrainSpec,
-    DriverUpgradePolicySpec,
-    WaitForCompletionSpec,
+        WaitForCompletionSpec,
 )
 from k8s_operator_libs_amd.core.client import FakeClient
 from k8s_operator_libs_amd.upgrade import consts, util

@@ -10,7 +10,7 @@ import pytest
 from k8s_operator_libs_amd.api.upgrade.v1alpha1 import DrainSpec
 from k8s_operator_libs_amd.core.errors import ApiError, NotFoundError
 from k8s_operator_libs_amd.upgrade import consts, util
-from k8s_operator_libs_amd.upgrade.drain import DrainError, drain_node
+from k8s_operator_libs_amd.upgrade.drain import drain_node
 from k8s_operator_libs_amd.upgrade.drain_manager import DrainConfiguration, DrainManager
 from k8s_operator_libs_amd.upgrade.node_state_provider import NodeUpgradeStateProvider
 

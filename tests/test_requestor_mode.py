@@ -4,7 +4,6 @@ upgrade_requestor.go)."""
 
 import pytest
 
-from k8s_operator_libs_amd.api.upgrade.v1alpha1 import DriverUpgradePolicySpec
 from k8s_operator_libs_amd.core.errors import NotFoundError
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.requestor import (

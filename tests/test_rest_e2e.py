@@ -11,7 +11,7 @@ from k8s_operator_libs_amd.core.apiserver import start_apiserver
 from k8s_operator_libs_amd.core.errors import ConflictError, NotFoundError
 from k8s_operator_libs_amd.core.restclient import RestClient
 from k8s_operator_libs_amd.crdutil import CRD_OPERATION_APPLY, process_crds
-from k8s_operator_libs_amd.upgrade import consts, util
+from k8s_operator_libs_amd.upgrade import consts
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
 from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
 

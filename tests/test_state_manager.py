@@ -9,11 +9,9 @@ single-node end-to-end with a simulated DaemonSet controller."""
 import pytest
 
 from k8s_operator_libs_amd.api.upgrade.v1alpha1 import (
-    DrainSpec,
-    DriverUpgradePolicySpec,
+        DriverUpgradePolicySpec,
 )
 from k8s_operator_libs_amd.upgrade import consts, util
-from k8s_operator_libs_amd.upgrade.common_manager import ClusterUpgradeState, NodeUpgradeState
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
 from k8s_operator_libs_amd.upgrade.state_manager import (
     BuildStateError,
