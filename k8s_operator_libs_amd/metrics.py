@@ -20,8 +20,16 @@ from typing import Dict, List, Optional, Tuple
 
 
 class Histogram:
-    """Fixed-bucket histogram with an exact-quantile reservoir (last 4096
-    observations) so reconcile p50/p99 can be reported precisely."""
+    """Fixed-bucket histogram with a uniform sampling reservoir.
+
+    Quantiles are EXACT while the observation count fits the reservoir
+    (4096) and thereafter come from Vitter's Algorithm R — a uniform random
+    sample over the ENTIRE run, so soak-length p99s are run-global rather
+    than window-local (the round-1 reservoir kept only
+    the most recent observations).  The fixed buckets are always run-global
+    and exact, and serve as the Prometheus exposition."""
+
+    RESERVOIR_SIZE = 4096
 
     DEFAULT_BUCKETS = (
         0.0001, 0.00025, 0.0005, 0.001, 0.0025, 0.005, 0.01, 0.025,
@@ -29,6 +37,8 @@ class Histogram:
     )
 
     def __init__(self, name: str, help_: str = "", buckets: Tuple[float, ...] = DEFAULT_BUCKETS):
+        import random
+
         self.name = name
         self.help = help_
         self.buckets = buckets
@@ -36,6 +46,8 @@ class Histogram:
         self._sum = 0.0
         self._n = 0
         self._reservoir: List[float] = []
+        # seeded per-instance: deterministic test runs, independent streams
+        self._rng = random.Random(0xA5A5 ^ hash(name))
         self._lock = threading.Lock()
 
     def observe(self, value: float) -> None:
@@ -44,12 +56,18 @@ class Histogram:
             self._counts[i] += 1
             self._sum += value
             self._n += 1
-            self._reservoir.append(value)
-            if len(self._reservoir) > 4096:
-                self._reservoir = self._reservoir[-2048:]
+            if len(self._reservoir) < self.RESERVOIR_SIZE:
+                self._reservoir.append(value)
+            else:
+                # Algorithm R: keep each of the n observations with equal
+                # probability RESERVOIR_SIZE/n
+                j = self._rng.randrange(self._n)
+                if j < self.RESERVOIR_SIZE:
+                    self._reservoir[j] = value
 
     def samples(self) -> List[float]:
-        """Recent raw observations (up to the reservoir size)."""
+        """Raw observations: exact when count <= RESERVOIR_SIZE, else a
+        uniform random sample over the whole run."""
         with self._lock:
             return list(self._reservoir)
 

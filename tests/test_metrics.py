@@ -103,3 +103,36 @@ def test_node_state_gauge_zeroes_on_transition(client):
     manager.build_state(DRIVER_NS, DRIVER_LABELS)
     assert reg.node_states.value(consts.UPGRADE_STATE_DRAIN_REQUIRED) == 0
     assert reg.node_states.value(consts.UPGRADE_STATE_DONE) == 1
+
+
+def test_histogram_quantiles_are_run_global():
+    """quantiles must reflect the WHOLE run, not the
+    last window.  A run whose large outliers all arrive in the first half
+    (pre-overflow) must still show them in p99 after 100k later small
+    observations displace the window a truncating reservoir would keep."""
+    from k8s_operator_libs_amd.metrics import Histogram
+
+    h = Histogram("t")
+    # 5% of the first 10k observations are 1.0s outliers...
+    for i in range(10_000):
+        h.observe(1.0 if i % 20 == 0 else 0.001)
+    # ...then 100k fast ones (a tail-window reservoir now holds only these)
+    for _ in range(100_000):
+        h.observe(0.001)
+    # run-global outlier fraction is 500/110000 ~ 0.45% (expected ~18.6 of
+    # the 4096 reservoir slots): the early outliers must still be visible
+    assert max(h.samples()) == 1.0
+    assert h.quantile(0.999) == 1.0
+    # and p50 stays at the common value
+    assert h.quantile(0.5) == 0.001
+    assert h.count == 110_000
+
+
+def test_histogram_exact_below_reservoir_capacity():
+    from k8s_operator_libs_amd.metrics import Histogram
+
+    h = Histogram("t2")
+    for v in [0.001, 0.002, 0.003, 0.004, 0.005]:
+        h.observe(v)
+    assert sorted(h.samples()) == [0.001, 0.002, 0.003, 0.004, 0.005]
+    assert h.quantile(0.5) == 0.003
