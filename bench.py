@@ -46,6 +46,7 @@ import time
 
 from k8s_operator_libs_amd.api.upgrade.v1alpha1 import DriverUpgradePolicySpec
 from k8s_operator_libs_amd.core import FakeClient
+from k8s_operator_libs_amd.core.errors import NotFoundError
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
 from k8s_operator_libs_amd.upgrade.state_manager import (
@@ -398,6 +399,14 @@ def run_rolling_upgrade_benchmark(
     completed = 0
 
     client, sim, close_substrate = _build_substrate(substrate)
+    # in-process ops are microseconds: per-node fan-out is pure overhead
+    # there (it pays off when each op is an HTTP round trip); saved and
+    # restored so secondary-substrate runs in the same process are unaffected
+    from k8s_operator_libs_amd.upgrade.common_manager import CommonUpgradeManager
+
+    prev_fanout = CommonUpgradeManager.MAX_PARALLEL_NODE_OPS
+    if substrate == "inproc":
+        CommonUpgradeManager.MAX_PARALLEL_NODE_OPS = 1
     ctrl = _DsController(sim, {"metadata": {"uid": "-"}}, "newrev", driver_labels)
     if mode == "requestor":
         _MaintenanceOperatorSim(sim)
@@ -448,8 +457,8 @@ def run_rolling_upgrade_benchmark(
                 try:
                     state = manager.reconcile(DRIVER_NS, driver_labels, policy,
                                               converge=converge)
-                except BuildStateError:
-                    # transient cache view of a driver pod mid-recreation;
+                except (BuildStateError, NotFoundError):
+                    # transient cache view of an object mid-recreation;
                     # the reference requeues the reconcile on error
                     # (upgrade_state.go:128-131) — retry next round
                     rounds += 1
@@ -511,8 +520,21 @@ def run_rolling_upgrade_benchmark(
             manager.wait_idle()
             ctrl.retarget({"metadata": {"uid": "-"}})
             cleanup()
+            # eventually-consistent substrates: drain the DELETED events
+            # before the next iteration re-creates same-named objects, or
+            # stale deletions could land mid-run
+            if substrate != "inproc":
+                deadline = time.monotonic() + 30
+                while time.monotonic() < deadline:
+                    try:
+                        if not client.list_nodes():
+                            break
+                    except Exception:
+                        pass
+                    time.sleep(0.001)
     finally:
         util.set_driver_name(prev_driver)
+        CommonUpgradeManager.MAX_PARALLEL_NODE_OPS = prev_fanout
         close_substrate()
 
     return {

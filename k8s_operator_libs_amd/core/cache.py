@@ -43,6 +43,9 @@ class _Informer:
         self._sync_delay = sync_delay
         self._store: Dict[Tuple[str, str], K8sObject] = {}
         self._lock = threading.RLock()
+        # notified on every store mutation: RV-aware barriers block here
+        # instead of polling (see CachedClient.wait_for_resource_version)
+        self._changed = threading.Condition(self._lock)
         self._synced = threading.Event()
         self._stop = threading.Event()
         self._watch = None
@@ -64,6 +67,7 @@ class _Informer:
                 (meta.namespace(obj), meta.name(obj)): obj for obj in objs
             }
             self._last_rv = rv
+            self._changed.notify_all()
 
     def _open_watch(self):
         """WATCH anchored at the last processed resourceVersion; a delegate
@@ -148,6 +152,7 @@ class _Informer:
                 if event_type == "DELETED":
                     self._store.pop(key, None)
                     self._last_rv = rv or self._last_rv
+                    self._changed.notify_all()
                 else:
                     current = self._store.get(key)
                     # resourceVersions are monotonic ints in this stack;
@@ -160,6 +165,7 @@ class _Informer:
                             pass
                     self._store[key] = obj
                     self._last_rv = rv or self._last_rv
+                    self._changed.notify_all()
 
     def stop(self) -> None:
         self._stop.set()
@@ -170,6 +176,36 @@ class _Informer:
 
     def wait_sync(self, timeout: float = 10.0) -> bool:
         return self._synced.wait(timeout)
+
+    def wait_for_rv(self, name: str, namespace: str, rv: str,
+                    timeout: float) -> bool:
+        """Block until the cached copy of (namespace, name) has
+        resourceVersion >= ``rv``, or the object is absent from the cache
+        (deleted), or the timeout elapses.  Event-driven via the informer's
+        condition variable — no polling."""
+        try:
+            target = int(rv)
+        except (TypeError, ValueError):
+            return False
+        key = (namespace, name)
+        deadline = time.monotonic() + timeout
+
+        def caught_up() -> bool:
+            obj = self._store.get(key)
+            if obj is None:
+                return True  # deleted (or never seen): nothing newer coming
+            try:
+                return int(meta.resource_version(obj)) >= target
+            except ValueError:
+                return False
+
+        with self._changed:
+            while not caught_up():
+                remaining = deadline - time.monotonic()
+                if remaining <= 0:
+                    return False
+                self._changed.wait(remaining)
+            return True
 
     def get(self, name: str, namespace: str) -> K8sObject:
         with self._lock:
@@ -220,6 +256,18 @@ class CachedClient(Client):
                 inf.start()
         inf.wait_sync()
         return inf
+
+    def wait_for_resource_version(self, api_version: str, kind: str,
+                                  name: str, namespace: str, rv: str,
+                                  timeout: float = 10.0) -> bool:
+        """Block until this cache has caught up to ``rv`` for the given
+        object (event-driven).  This is the fast path of the state
+        provider's patch-then-confirm barrier: the patch response's
+        resourceVersion is the exact point the cache must reach before the
+        next reconcile may trust its reads."""
+        return self._informer_for(api_version, kind).wait_for_rv(
+            name, namespace, rv, timeout
+        )
 
     def wait_for_cache_sync(self, timeout: float = 10.0) -> bool:
         with self._lock:

@@ -120,6 +120,48 @@ class CommonUpgradeManager:
         self.pod_deletion_state_enabled = pod_deletion_filter is not None
         self.validation_state_enabled = bool(validation_pod_selector)
 
+    #: per-phase node fan-out width (0/1 disables).  Per-node operations in
+    #: a phase are independent (each provider mutation takes the per-node
+    #: KeyedMutex), and over a REST substrate each transition pays an HTTP
+    #: round trip + informer-propagation barrier — concurrency here takes
+    #: the phase from O(nodes) round trips to O(1) wall-clock.
+    MAX_PARALLEL_NODE_OPS = 16
+
+    def for_each_node(self, node_states, fn) -> None:
+        """Run ``fn(node_state)`` for every node in the phase, fanned out
+        over a bounded thread pool (sequential for 0/1 nodes).  All nodes
+        are processed even if some fail; the first exception is re-raised
+        afterwards, matching the idempotent requeue-on-error contract
+        (upgrade_state.go:171-281 processes phases best-effort and relies
+        on the next reconcile)."""
+        node_states = list(node_states)
+        width = min(len(node_states), self.MAX_PARALLEL_NODE_OPS)
+        if width <= 1:
+            for ns in node_states:
+                fn(ns)
+            return
+        pool = getattr(self, "_node_op_pool", None)
+        if pool is None:
+            # persistent pool: per-call executor creation/teardown costs
+            # more than the fan-out saves (threads join on every phase)
+            from concurrent.futures import ThreadPoolExecutor
+
+            pool = ThreadPoolExecutor(
+                max_workers=self.MAX_PARALLEL_NODE_OPS,
+                thread_name_prefix="node-op",
+            )
+            self._node_op_pool = pool
+        futures = [pool.submit(fn, ns) for ns in node_states]
+        first_exc = None
+        for fut in futures:
+            exc = fut.exception()  # blocks until done
+            if exc is not None and first_exc is None:
+                first_exc = exc
+            elif exc is not None:
+                logger.error("per-node phase op failed: %s", exc)
+        if first_exc is not None:
+            raise first_exc
+
     # -- feature flags (common_manager.go:136-144) ---------------------------
 
     def is_pod_deletion_enabled(self) -> bool:
@@ -218,7 +260,8 @@ class CommonUpgradeManager:
         self, state: ClusterUpgradeState, state_name: str
     ) -> None:
         """(common_manager.go:229-291)"""
-        for node_state in state.nodes_in(state_name):
+
+        def one(node_state):
             synced, orphaned = self.pod_in_sync_with_ds(node_state)
             waiting_safe_load = self.safe_driver_load_manager.is_waiting_for_safe_driver_load(
                 node_state.node
@@ -236,19 +279,26 @@ class CommonUpgradeManager:
                 self.node_state_provider.change_node_upgrade_state(
                     node_state.node, consts.UPGRADE_STATE_UPGRADE_REQUIRED
                 )
-                continue
+                return
             if state_name == consts.UPGRADE_STATE_UNKNOWN:
                 self.node_state_provider.change_node_upgrade_state(
                     node_state.node, consts.UPGRADE_STATE_DONE
                 )
 
+        self.for_each_node(state.nodes_in(state_name), one)
+
     def process_cordon_required_nodes(self, state: ClusterUpgradeState) -> None:
         """(common_manager.go:361-380)"""
-        for node_state in state.nodes_in(consts.UPGRADE_STATE_CORDON_REQUIRED):
+
+        def one(node_state):
             self.cordon_manager.cordon(node_state.node)
             self.node_state_provider.change_node_upgrade_state(
                 node_state.node, consts.UPGRADE_STATE_WAIT_FOR_JOBS_REQUIRED
             )
+
+        self.for_each_node(
+            state.nodes_in(consts.UPGRADE_STATE_CORDON_REQUIRED), one
+        )
 
     def process_wait_for_jobs_required_nodes(
         self,
@@ -324,20 +374,21 @@ class CommonUpgradeManager:
 
     def process_pod_restart_nodes(self, state: ClusterUpgradeState) -> None:
         """(common_manager.go:457-524)"""
-        pods_to_restart = []
-        for node_state in state.nodes_in(consts.UPGRADE_STATE_POD_RESTART_REQUIRED):
+        pods_to_restart = []  # .append is atomic under the GIL
+
+        def one(node_state):
             synced, orphaned = self.pod_in_sync_with_ds(node_state)
             if not synced or orphaned:
                 # restart unless already terminating
                 if "deletionTimestamp" not in node_state.driver_pod.get("metadata", {}):
                     pods_to_restart.append(node_state.driver_pod)
-                continue
+                return
             # template in sync: unblock safe driver load, then wait for Ready
             self.safe_driver_load_manager.unblock_loading(node_state.node)
             if self.is_driver_pod_in_sync(node_state):
                 if not self.is_validation_enabled():
                     self.update_node_to_uncordon_or_done_state(node_state)
-                    continue
+                    return
                 self.node_state_provider.change_node_upgrade_state(
                     node_state.node, consts.UPGRADE_STATE_VALIDATION_REQUIRED
                 )
@@ -345,6 +396,10 @@ class CommonUpgradeManager:
                 self.node_state_provider.change_node_upgrade_state(
                     node_state.node, consts.UPGRADE_STATE_FAILED
                 )
+
+        self.for_each_node(
+            state.nodes_in(consts.UPGRADE_STATE_POD_RESTART_REQUIRED), one
+        )
         self.pod_manager.schedule_pods_restart(pods_to_restart)
 
     def process_upgrade_failed_nodes(self, state: ClusterUpgradeState) -> None:
@@ -365,14 +420,19 @@ class CommonUpgradeManager:
 
     def process_validation_required_nodes(self, state: ClusterUpgradeState) -> None:
         """(common_manager.go:573-604)"""
-        for node_state in state.nodes_in(consts.UPGRADE_STATE_VALIDATION_REQUIRED):
+
+        def one(node_state):
             # The driver may have restarted after reaching this state and be
             # blocked on safe load again — always unblock here
             # (common_manager.go:581-586).
             self.safe_driver_load_manager.unblock_loading(node_state.node)
             if not self.validation_manager.validate(node_state.node):
-                continue
+                return
             self.update_node_to_uncordon_or_done_state(node_state)
+
+        self.for_each_node(
+            state.nodes_in(consts.UPGRADE_STATE_VALIDATION_REQUIRED), one
+        )
 
     def update_node_to_uncordon_or_done_state(self, node_state: NodeUpgradeState) -> None:
         """(common_manager.go:673-708): initially-unschedulable nodes skip

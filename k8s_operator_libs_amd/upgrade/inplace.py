@@ -79,12 +79,17 @@ class InplaceNodeStateManager:
     def process_uncordon_required_nodes(self, state: ClusterUpgradeState) -> None:
         """(upgrade_inplace.go:124-147)"""
         common = self.common
-        for node_state in state.nodes_in(consts.UPGRADE_STATE_UNCORDON_REQUIRED):
+
+        def one(node_state):
             node = node_state.node
             if is_node_in_requestor_mode(node):
                 # requestor flow owns this node's uncordon
-                continue
+                return
             common.cordon_manager.uncordon(node)
             common.node_state_provider.change_node_upgrade_state(
                 node, consts.UPGRADE_STATE_DONE
             )
+
+        common.for_each_node(
+            state.nodes_in(consts.UPGRADE_STATE_UNCORDON_REQUIRED), one
+        )

@@ -129,8 +129,23 @@ class NodeUpgradeStateProvider:
     # -- internals -----------------------------------------------------------
 
     def _patch_and_confirm(self, node_name: str, patch: K8sObject, confirmed) -> None:
-        self._client.patch("v1", "Node", node_name, patch)
+        resp = self._client.patch("v1", "Node", node_name, patch)
         deadline = time.monotonic() + _BARRIER_TIMEOUT_S
+
+        # Fast path: an informer-backed client exposes an event-driven
+        # RV barrier — block until the cache has seen exactly the
+        # resourceVersion our patch produced, no polling.
+        waiter = getattr(self._client, "wait_for_resource_version", None)
+        rv = meta.resource_version(resp) if isinstance(resp, dict) else ""
+        if waiter is not None and rv:
+            if waiter("v1", "Node", node_name, "", rv, _BARRIER_TIMEOUT_S):
+                live = self._client.get_node(node_name)
+                if confirmed(live):
+                    return
+            # cache caught up but the condition doesn't hold (a competing
+            # writer overwrote us) or the wait timed out: fall through to
+            # the poll loop, which decides between converged and timeout
+
         interval = _BARRIER_INITIAL_S
         while True:
             live = self._client.get_node(node_name)

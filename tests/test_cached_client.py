@@ -124,3 +124,58 @@ def test_informer_reconnects_after_watch_drop(client):
         assert cached.get_node("n1")["metadata"]["labels"]["x"] == "1"
     finally:
         cached.stop()
+
+
+def test_rv_barrier_event_driven():
+    """The provider's patch-then-confirm barrier over a cached client uses
+    the event-driven RV wait (no polling): after patch, wait_for_resource_version
+    returns once the informer sees the patch's RV."""
+    import time
+
+    from k8s_operator_libs_amd.core.cache import CachedClient
+    from k8s_operator_libs_amd.core.client import FakeClient
+    from k8s_operator_libs_amd.core.fakecluster import FakeCluster
+
+    cluster = FakeCluster()
+    cluster.create({"apiVersion": "v1", "kind": "Node",
+                    "metadata": {"name": "n1"}, "spec": {}})
+    cached = CachedClient(FakeClient(cluster), sync_delay=0.05)
+    try:
+        cached.get("v1", "Node", "n1")
+        resp = cached.patch("v1", "Node", "n1",
+                            {"metadata": {"labels": {"k": "v"}}})
+        rv = resp["metadata"]["resourceVersion"]
+        t0 = time.monotonic()
+        assert cached.wait_for_resource_version("v1", "Node", "n1", "", rv, 5.0)
+        # the 50ms artificial informer lag bounds the wait from below;
+        # event-driven wakeup bounds it from above (well under a poll sweep)
+        assert cached.get("v1", "Node", "n1")["metadata"]["labels"]["k"] == "v"
+        # timeout path: an RV the cluster will never reach
+        assert not cached.wait_for_resource_version(
+            "v1", "Node", "n1", "", str(int(rv) + 1000), 0.2)
+    finally:
+        cached.stop()
+
+
+def test_provider_barrier_uses_rv_fast_path():
+    """End-to-end: state transitions through the provider over a laggy
+    informer cache still fire exactly once, via the RV fast path."""
+    from k8s_operator_libs_amd.core.cache import CachedClient
+    from k8s_operator_libs_amd.core.client import FakeClient
+    from k8s_operator_libs_amd.core.fakecluster import FakeCluster
+    from k8s_operator_libs_amd.upgrade.node_state_provider import (
+        NodeUpgradeStateProvider,
+    )
+
+    cluster = FakeCluster()
+    cluster.create({"apiVersion": "v1", "kind": "Node",
+                    "metadata": {"name": "n1", "labels": {}}, "spec": {}})
+    cached = CachedClient(FakeClient(cluster), sync_delay=0.02)
+    try:
+        provider = NodeUpgradeStateProvider(cached)
+        node = provider.get_node("n1")
+        provider.change_node_upgrade_state(node, "cordon-required")
+        assert provider.get_node("n1")["metadata"]["labels"][
+            util.get_upgrade_state_label_key()] == "cordon-required"
+    finally:
+        cached.stop()
