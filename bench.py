@@ -420,6 +420,11 @@ def run_rolling_upgrade_benchmark(
             # eventually-consistent substrates: wait until the client view
             # reflects the fully-built cluster before the timed region (the
             # production analogue is controller-runtime WaitForCacheSync)
+            # the client must see the COMPLETE object set before the timed
+            # region: a partial view (e.g. train pods not yet cached) makes
+            # the eviction count mismatch and costs extra state hops
+            want_default_ns = n_nodes * gpu_pods_per_node + (
+                n_nodes if mode == "anic" else 0)
             deadline = time.monotonic() + 30
             while time.monotonic() < deadline:
                 try:
@@ -427,11 +432,18 @@ def run_rolling_upgrade_benchmark(
                                                  label_selector=",".join(
                                                      f"{k}={v}" for k, v in
                                                      driver_labels.items()))
-                    pods = client.list_pods(namespace=DRIVER_NS)
-                    if dss and len([p for p in pods
+                    driver_ns_pods = client.list_pods(namespace=DRIVER_NS)
+                    n_driver = len([p for p in driver_ns_pods
                                     if p["metadata"]["labels"].get(
-                                        "controller-revision-hash")]) >= n_nodes \
-                            and len(client.list_nodes()) >= n_nodes:
+                                        "controller-revision-hash")])
+                    n_validator = len([p for p in driver_ns_pods
+                                       if p["metadata"]["labels"].get("app")
+                                       == "amd-gpu-validator"])
+                    n_workload = len(client.list_pods(namespace="default"))
+                    if (dss and n_driver >= n_nodes
+                            and n_validator >= n_nodes
+                            and n_workload >= want_default_ns
+                            and len(client.list_nodes()) >= n_nodes):
                         break
                 except Exception:
                     pass
@@ -500,7 +512,18 @@ def run_rolling_upgrade_benchmark(
                     == consts.UPGRADE_STATE_DONE
                 )
                 if done == n_nodes:
-                    break
+                    # measurement integrity: completion must hold in the
+                    # AUTHORITATIVE store, not just this client's (possibly
+                    # stale) view — round-1's cached numbers broke early on
+                    # the previous iteration's done labels still in cache
+                    sim_done = sum(
+                        1 for n in sim.list("v1", "Node")
+                        if (n["metadata"].get("labels") or {}).get(state_key)
+                        == consts.UPGRADE_STATE_DONE
+                    )
+                    if sim_done == n_nodes:
+                        break
+                    done = -1  # view was stale: keep reconciling
                 if rounds > 200 * n_nodes:
                     raise RuntimeError(
                         f"upgrade did not converge after {rounds} rounds"
@@ -520,9 +543,8 @@ def run_rolling_upgrade_benchmark(
             manager.wait_idle()
             ctrl.retarget({"metadata": {"uid": "-"}})
             cleanup()
-            # eventually-consistent substrates: drain the DELETED events
-            # before the next iteration re-creates same-named objects, or
-            # stale deletions could land mid-run
+            # iteration isolation: drain the previous cluster's DELETED
+            # events out of the cache before re-creating same-named objects
             if substrate != "inproc":
                 deadline = time.monotonic() + 30
                 while time.monotonic() < deadline:

@@ -120,12 +120,17 @@ class CommonUpgradeManager:
         self.pod_deletion_state_enabled = pod_deletion_filter is not None
         self.validation_state_enabled = bool(validation_pod_selector)
 
-    #: per-phase node fan-out width (0/1 disables).  Per-node operations in
-    #: a phase are independent (each provider mutation takes the per-node
-    #: KeyedMutex), and over a REST substrate each transition pays an HTTP
-    #: round trip + informer-propagation barrier — concurrency here takes
-    #: the phase from O(nodes) round trips to O(1) wall-clock.
-    MAX_PARALLEL_NODE_OPS = 16
+    #: per-phase node fan-out width (0/1 = sequential, the default).
+    #: Per-node operations in a phase are independent (each provider
+    #: mutation takes the per-node KeyedMutex), so against a REMOTE
+    #: apiserver — where each transition pays a genuine network round
+    #: trip — raising this takes a phase from O(nodes) round trips to
+    #: O(1) wall-clock.  It is OFF by default because it measured 2-3x
+    #: SLOWER on the benchmark topology (apiserver in the same process:
+    #: the "concurrent" requests contend for the GIL and the server's
+    #: event loop instead of overlapping wire latency).  Raise it only
+    #: for real out-of-process clusters, verified with your own numbers.
+    MAX_PARALLEL_NODE_OPS = 1
 
     def for_each_node(self, node_states, fn) -> None:
         """Run ``fn(node_state)`` for every node in the phase, fanned out
