@@ -347,23 +347,45 @@ def _wrap_nogroup(fn):
 class ApiServerHandle:
     """A running mini-apiserver on a background thread."""
 
-    def __init__(self, server, thread, cluster, url):
+    def __init__(self, server, thread, cluster, url, engine="thread"):
         self._server = server
         self._thread = thread
         self.cluster = cluster
         self.url = url
+        self.engine = engine
 
     def stop(self) -> None:
-        self._server.should_exit = True
+        if self.engine == "uvicorn":
+            self._server.should_exit = True
+        else:
+            self._server._shutting_down = True
+            self._server.shutdown()
+            self._server.server_close()
         self._thread.join(timeout=10)
 
 
 def start_apiserver(
     host: str = "127.0.0.1", port: int = 0,
     cluster: Optional[FakeCluster] = None,
+    engine: str = "thread",
 ) -> ApiServerHandle:
     """Start the mini-apiserver on a background thread; returns a handle with
-    the bound URL (port=0 picks a free port)."""
+    the bound URL (port=0 picks a free port).
+
+    ``engine="thread"`` (default) uses the lean stdlib threaded server —
+    the benchmark substrate; ``engine="uvicorn"`` keeps the FastAPI/asyncio
+    implementation for cross-checking the wire surface against a second
+    independent stack (tests/test_rest_e2e.py runs a matrix)."""
+    cluster = cluster or FakeCluster()
+
+    if engine == "thread":
+        server = _make_threaded_server(host, port, cluster)
+        thread = threading.Thread(target=server.serve_forever,
+                                  kwargs={"poll_interval": 0.05}, daemon=True)
+        thread.start()
+        url = f"http://{host}:{server.server_address[1]}"
+        return ApiServerHandle(server, thread, cluster, url, engine="thread")
+
     import socket
 
     import uvicorn
@@ -384,4 +406,258 @@ def start_apiserver(
         if time.monotonic() > deadline:
             raise RuntimeError("mini-apiserver failed to start in 15s")
         time.sleep(0.01)
-    return ApiServerHandle(server, thread, cluster, url)
+    return ApiServerHandle(server, thread, cluster, url, engine="uvicorn")
+
+
+# ---------------------------------------------------------------------------
+# threaded engine: a lean stdlib HTTP server for the same API surface
+# ---------------------------------------------------------------------------
+
+def _make_threaded_server(host, port, cluster):
+    """A hand-rolled threaded HTTP apiserver over the same FakeCluster.
+
+    Serves the identical wire surface as the FastAPI app (CRUD, merge
+    patch, selectors, pagination, status/eviction subresources, discovery,
+    RV-anchored watches with bookmarks/ERROR-410) but through
+    ``http.server`` with HTTP/1.1 keep-alive and no asyncio: per-request
+    overhead is several times lower than the uvicorn engine, which matters
+    because this server is the benchmark substrate (the reconcile loop's
+    wire cost is dominated by it).  Watch streams use read-until-close
+    framing on a dedicated connection."""
+    import http.server
+    import socketserver
+    import urllib.parse
+
+    from .errors import ApiError, GoneError
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"
+        # TCP_NODELAY: without it, Nagle + delayed-ACK interplay stalls
+        # small request/response pairs by ~40ms each
+        disable_nagle_algorithm = True
+
+        # -- plumbing --------------------------------------------------------
+
+        def log_message(self, fmt, *args):  # quiet
+            pass
+
+        def _send_json(self, obj, code=200):
+            body = json.dumps(obj).encode()
+            self.send_response(code)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def _read_body(self):
+            length = int(self.headers.get("Content-Length") or 0)
+            if not length:
+                return {}
+            return json.loads(self.rfile.read(length) or b"{}")
+
+        def _route(self):
+            """Parse the request path into
+            (api_version, plural, namespace, name, subresource, query)."""
+            parsed = urllib.parse.urlsplit(self.path)
+            qp = dict(urllib.parse.parse_qsl(parsed.query))
+            parts = [p for p in parsed.path.split("/") if p]
+            if not parts:
+                raise ApiError("not found")
+            if parts[0] == "api":
+                api_version = parts[1] if len(parts) > 1 else ""
+                rest = parts[2:]
+            elif parts[0] == "apis":
+                if len(parts) < 3:
+                    raise ApiError("not found")
+                api_version = f"{parts[1]}/{parts[2]}"
+                rest = parts[3:]
+            else:
+                raise ApiError("not found")
+            namespace = ""
+            if rest[:1] == ["namespaces"] and len(rest) >= 2:
+                # bare namespace-object ops go through the core route
+                if len(rest) == 2 and api_version == "v1":
+                    return api_version, "namespaces", "", rest[1], "", qp
+                namespace = rest[1]
+                rest = rest[2:]
+            plural = rest[0] if rest else ""
+            name = rest[1] if len(rest) > 1 else ""
+            sub = rest[2] if len(rest) > 2 else ""
+            return api_version, plural, namespace, name, sub, qp
+
+        def _resolve(self, api_version, plural):
+            kind = cluster.lookup_by_plural(api_version, plural)
+            if kind is None:
+                raise ApiError(f"resource {plural} not served in {api_version}")
+            return kind
+
+        def _handle(self, fn):
+            try:
+                fn()
+            except ApiError as exc:
+                self._send_json(_status_body(exc), code=exc.code)
+            except BrokenPipeError:
+                pass
+
+        # -- methods ---------------------------------------------------------
+
+        def do_GET(self):
+            def run():
+                api_version, plural, namespace, name, sub, qp = self._route()
+                if not plural:  # discovery
+                    group_version = api_version
+                    resources = []
+                    for (av, kind), (pl, namespaced) in cluster._kinds.items():
+                        if av == group_version:
+                            resources.append({"name": pl, "kind": kind,
+                                              "namespaced": namespaced})
+                    self._send_json({"kind": "APIResourceList",
+                                     "groupVersion": group_version,
+                                     "resources": resources})
+                    return
+                kind = self._resolve(api_version, plural)
+                if name:
+                    self._send_json(cluster.get(api_version, kind, name, namespace))
+                    return
+                if qp.get("watch") in ("true", "1"):
+                    self._stream_watch(api_version, kind, namespace, qp)
+                    return
+                try:
+                    limit = int(qp.get("limit", "0"))
+                except ValueError:
+                    limit = 0
+                items, rv, cont = cluster.list_paged(
+                    api_version, kind, namespace=namespace or None,
+                    label_selector=qp.get("labelSelector", ""),
+                    field_selector=qp.get("fieldSelector", ""),
+                    limit=limit, continue_token=qp.get("continue", ""),
+                )
+                md = {"resourceVersion": rv}
+                if cont:
+                    md["continue"] = cont
+                self._send_json({"kind": f"{kind}List",
+                                 "apiVersion": api_version,
+                                 "metadata": md, "items": items})
+            self._handle(run)
+
+        def _stream_watch(self, api_version, kind, namespace, qp):
+            from .errors import GoneError as _Gone
+
+            bookmarks = qp.get("allowWatchBookmarks") in ("true", "1")
+            try:
+                timeout_s = float(qp.get("timeoutSeconds", "0")) or None
+            except ValueError:
+                timeout_s = None
+            try:
+                watch = cluster.watch(
+                    api_version, kind, namespace=namespace or None,
+                    resource_version=qp.get("resourceVersion") or None,
+                    label_selector=qp.get("labelSelector", ""),
+                )
+            except _Gone as exc:
+                # 200 + in-stream ERROR Status, the real apiserver shape
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Connection", "close")
+                self.end_headers()
+                self.wfile.write((json.dumps({
+                    "type": "ERROR",
+                    "object": {"kind": "Status", "apiVersion": "v1",
+                               "status": "Failure", "reason": "Expired",
+                               "message": exc.message, "code": 410},
+                }) + "\n").encode())
+                self.close_connection = True
+                return
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Connection", "close")  # read-until-close framing
+            self.end_headers()
+            self.close_connection = True
+            deadline = time.monotonic() + timeout_s if timeout_s else None
+            last_bookmark = None
+            try:
+                while not getattr(server, "_shutting_down", False):
+                    if deadline is not None and time.monotonic() >= deadline:
+                        return
+                    item = watch.next(timeout=0.5)
+                    if item is None:
+                        if bookmarks:
+                            rv = watch.bookmark_rv()
+                            if rv is not None and rv != last_bookmark:
+                                last_bookmark = rv
+                                self.wfile.write((json.dumps({
+                                    "type": "BOOKMARK",
+                                    "object": {"kind": kind,
+                                               "apiVersion": api_version,
+                                               "metadata": {"resourceVersion": rv}},
+                                }) + "\n").encode())
+                                self.wfile.flush()
+                                continue
+                        # keep-alive probe surfaces dead clients
+                        self.wfile.write(b"\n")
+                        self.wfile.flush()
+                        continue
+                    event_type, obj = item
+                    self.wfile.write((json.dumps(
+                        {"type": event_type, "object": obj}) + "\n").encode())
+                    self.wfile.flush()
+            except (BrokenPipeError, ConnectionResetError, OSError):
+                pass
+            finally:
+                watch.stop()
+
+        def do_POST(self):
+            def run():
+                api_version, plural, namespace, name, sub, qp = self._route()
+                kind = self._resolve(api_version, plural)
+                body = self._read_body()
+                if sub == "eviction" and kind == "Pod":
+                    cluster.evict_pod(name, namespace)
+                    self._send_json({"kind": "Status", "status": "Success"})
+                    return
+                if namespace:
+                    body.setdefault("metadata", {})["namespace"] = namespace
+                self._send_json(cluster.create(body), code=201)
+            self._handle(run)
+
+        def do_PUT(self):
+            def run():
+                api_version, plural, namespace, name, sub, qp = self._route()
+                kind = self._resolve(api_version, plural)
+                body = self._read_body()
+                if sub == "status":
+                    self._send_json(cluster.patch(
+                        api_version, kind, name,
+                        {"status": body.get("status", body)}, namespace))
+                    return
+                if namespace:
+                    body.setdefault("metadata", {})["namespace"] = namespace
+                self._send_json(cluster.update(body))
+            self._handle(run)
+
+        def do_PATCH(self):
+            def run():
+                api_version, plural, namespace, name, sub, qp = self._route()
+                kind = self._resolve(api_version, plural)
+                body = self._read_body()
+                if sub == "status":
+                    body = {"status": body.get("status", body)}
+                self._send_json(cluster.patch(api_version, kind, name, body,
+                                              namespace))
+            self._handle(run)
+
+        def do_DELETE(self):
+            def run():
+                api_version, plural, namespace, name, sub, qp = self._route()
+                kind = self._resolve(api_version, plural)
+                self._read_body()  # DeleteOptions accepted, grace ignored here
+                cluster.delete(api_version, kind, name, namespace)
+                self._send_json({"kind": "Status", "status": "Success"})
+            self._handle(run)
+
+    class Server(socketserver.ThreadingMixIn, http.server.HTTPServer):
+        daemon_threads = True
+        allow_reuse_address = True
+
+    server = Server((host, port), Handler)
+    return server

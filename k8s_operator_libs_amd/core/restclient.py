@@ -79,13 +79,64 @@ class RestClient(Client):
         headers = {"Content-Type": "application/json"}
         if token:
             headers["Authorization"] = f"Bearer {token}"
+        self._headers = dict(headers)
+        self._timeout = timeout
         self._http = httpx.Client(
             base_url=self.base_url, headers=headers, verify=verify, timeout=timeout
         )
+        # Plain-HTTP fast path: thread-local persistent http.client
+        # connections cut ~40% off per-request latency vs the full httpx
+        # stack — and request latency IS the reconcile loop's wire cost.
+        # TLS endpoints (real apiservers) stay on httpx.
+        self._fast_netloc = None
+        if self.base_url.startswith("http://"):
+            import urllib.parse
+
+            parsed = urllib.parse.urlsplit(self.base_url)
+            self._fast_netloc = (parsed.hostname, parsed.port or 80)
+            self._fast_local = __import__("threading").local()
         self._kinds = dict(_KIND_INFO)
         if extra_kinds:
             self._kinds.update(extra_kinds)
         self._retries = retries
+
+    # -- plain-HTTP fast transport -------------------------------------------
+
+    def _fast_request(self, method: str, path: str, params=None, content=None,
+                      headers=None):
+        import http.client
+        import urllib.parse
+
+        if params:
+            path = f"{path}?{urllib.parse.urlencode(params)}"
+        conn = getattr(self._fast_local, "conn", None)
+        hdrs = dict(self._headers)
+        if headers:
+            hdrs.update(headers)
+        body = content if content is not None else None
+        for attempt in (0, 1):
+            if conn is None:
+                conn = http.client.HTTPConnection(
+                    *self._fast_netloc, timeout=self._timeout
+                )
+                self._fast_local.conn = conn
+            try:
+                conn.request(method, path, body=body, headers=hdrs)
+                resp = conn.getresponse()
+                data = resp.read()
+                return _FastResponse(resp.status, data)
+            except (http.client.HTTPException, ConnectionError, OSError):
+                # keep-alive connection died (server restart/idle close):
+                # reconnect once, then let the caller's retry logic own it
+                try:
+                    conn.close()
+                except Exception:
+                    pass
+                conn = None
+                self._fast_local.conn = None
+                if attempt:
+                    raise
+        raise AssertionError("unreachable")
 
     def _request(self, method: str, path: str, **kw) -> httpx.Response:
         """Issue a request with client-go-style retries on transient failures
@@ -96,9 +147,13 @@ class RestClient(Client):
         attempt = 0
         while True:
             try:
-                resp = self._http.request(method, path, **kw)
+                if self._fast_netloc is not None:
+                    resp = self._fast_request(method, path, **kw)
+                else:
+                    resp = self._http.request(method, path, **kw)
             except (httpx.ConnectError, httpx.ReadError, httpx.RemoteProtocolError,
-                    httpx.ConnectTimeout, httpx.ReadTimeout) as exc:
+                    httpx.ConnectTimeout, httpx.ReadTimeout,
+                    ConnectionError, OSError) as exc:
                 if attempt >= self._retries:
                     err = ApiError(f"connection to apiserver failed: {exc}")
                     err.code = 503
@@ -342,6 +397,24 @@ class RestClient(Client):
 
     def close(self) -> None:
         self._http.close()
+
+
+class _FastResponse:
+    """Minimal response shim matching the httpx surface `_raise_for` and
+    the CRUD methods consume (status_code / json() / text)."""
+
+    __slots__ = ("status_code", "_data")
+
+    def __init__(self, status_code: int, data: bytes) -> None:
+        self.status_code = status_code
+        self._data = data
+
+    def json(self):
+        return json.loads(self._data)
+
+    @property
+    def text(self) -> str:
+        return self._data.decode(errors="replace")
 
 
 class _HttpWatch:
