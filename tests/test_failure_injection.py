@@ -189,3 +189,89 @@ class TestAnicNicDriverFlow:
                          {"name": "main", "ready": True, "restartCount": 0}]}},
                      "default")
         assert run_until(client, manager, pol, consts.UPGRADE_STATE_DONE)
+
+
+class TestApiserverOutage:
+    """The apiserver itself restarts mid-upgrade: the informer stack must
+    reconnect (re-watch from last RV / relist), the keep-alive fast
+    transport must re-establish connections, and the upgrade must complete
+    without double-fired transitions.  The reference inherits this from
+    client-go; here it is pinned wire-level."""
+
+    def test_upgrade_survives_apiserver_restart(self):
+        import socket
+        import time
+
+        from k8s_operator_libs_amd.core.apiserver import start_apiserver
+        from k8s_operator_libs_amd.core.cache import CachedClient
+        from k8s_operator_libs_amd.core.fakecluster import FakeCluster
+        from k8s_operator_libs_amd.core.restclient import RestClient
+        from k8s_operator_libs_amd.metrics import MetricsRegistry
+        from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
+        from k8s_operator_libs_amd.upgrade.state_manager import (
+            BuildStateError,
+            ClusterUpgradeStateManager,
+        )
+        from k8s_operator_libs_amd.core.errors import ApiError, NotFoundError
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+
+        cluster = FakeCluster()  # state survives the server process
+        handle = start_apiserver(port=port, cluster=cluster)
+        rest = RestClient(handle.url)
+        cached = CachedClient(rest)
+
+        class W:
+            pass
+
+        W.cluster = cluster
+        ds, _ = setup_cluster(W, n_nodes=2, pod_hash="old", ds_hash="new")
+        SimDaemonSetController(cluster, ds, current_hash="new")
+
+        registry = MetricsRegistry()
+        manager = ClusterUpgradeStateManager(cached, metrics=registry) \
+            .with_pod_deletion_enabled(gpu_pod_deletion_filter)
+        pol = policy(maxParallelUpgrades=1, maxUnavailable="100%",
+                     drainSpec={"enable": True})
+
+        restarted = False
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            try:
+                manager.reconcile(DRIVER_NS, DRIVER_LABELS, pol)
+            except (BuildStateError, NotFoundError, ApiError):
+                time.sleep(0.05)
+                continue
+            states = {
+                n["metadata"]["labels"].get(util.get_upgrade_state_label_key())
+                for n in cluster.list("v1", "Node")
+            }
+            if not restarted and states & {"pod-restart-required",
+                                           "drain-required",
+                                           "pod-deletion-required"}:
+                # outage in the middle of the pipeline
+                handle.stop()
+                time.sleep(0.2)
+                handle = start_apiserver(port=port, cluster=cluster)
+                restarted = True
+            if states == {consts.UPGRADE_STATE_DONE}:
+                break
+            time.sleep(0.05)
+
+        try:
+            assert restarted, "restart point never reached"
+            node_states = [
+                n["metadata"]["labels"].get(util.get_upgrade_state_label_key())
+                for n in cluster.list("v1", "Node")
+            ]
+            assert node_states == [consts.UPGRADE_STATE_DONE] * 2, node_states
+            # no transition double-fired despite the outage
+            for (frm, to), count in registry.state_transitions.items().items():
+                assert count <= 2, f"{frm}->{to} fired {count} times"
+        finally:
+            manager.wait_idle()
+            cached.stop()
+            rest.close()
+            handle.stop()
