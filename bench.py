@@ -49,6 +49,7 @@ from k8s_operator_libs_amd.core import FakeClient
 from k8s_operator_libs_amd.core.errors import NotFoundError
 from k8s_operator_libs_amd.upgrade import consts, util
 from k8s_operator_libs_amd.upgrade.drain import gpu_pod_deletion_filter
+from k8s_operator_libs_amd.upgrade.pod_manager import StaleClusterViewError
 from k8s_operator_libs_amd.upgrade.state_manager import (
     BuildStateError,
     ClusterUpgradeStateManager,
@@ -440,7 +441,9 @@ def run_rolling_upgrade_benchmark(
                                        if p["metadata"]["labels"].get("app")
                                        == "amd-gpu-validator"])
                     n_workload = len(client.list_pods(namespace="default"))
-                    if (dss and n_driver >= n_nodes
+                    n_revs = len(client.list_controller_revisions(
+                        namespace=DRIVER_NS))
+                    if (dss and n_revs >= 2 and n_driver >= n_nodes
                             and n_validator >= n_nodes
                             and n_workload >= want_default_ns
                             and len(client.list_nodes()) >= n_nodes):
@@ -469,7 +472,7 @@ def run_rolling_upgrade_benchmark(
                 try:
                     state = manager.reconcile(DRIVER_NS, driver_labels, policy,
                                               converge=converge)
-                except (BuildStateError, NotFoundError):
+                except (BuildStateError, NotFoundError, StaleClusterViewError):
                     # transient cache view of an object mid-recreation;
                     # the reference requeues the reconcile on error
                     # (upgrade_state.go:128-131) — retry next round

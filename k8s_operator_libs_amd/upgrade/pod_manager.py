@@ -45,6 +45,14 @@ from .node_state_provider import NodeUpgradeStateProvider
 logger = logging.getLogger(__name__)
 
 # Pod label set by the DaemonSet controller with the ControllerRevision hash.
+class StaleClusterViewError(RuntimeError):
+    """The client's view is missing objects that must exist (e.g. a managed
+    DaemonSet with no ControllerRevisions): with an informer-backed client
+    this is a transient staleness window, not a cluster defect.  Callers
+    treat it like any reconcile error — requeue (upgrade_state.go:128-131
+    pattern)."""
+
+
 POD_CONTROLLER_REVISION_HASH_LABEL = "controller-revision-hash"
 
 PodDeletionFilter = Callable[[K8sObject], bool]
@@ -108,7 +116,7 @@ class PodManager:
             namespace=meta.namespace(daemonset), label_selector=label_selector
         )
         if not revisions:
-            raise ValueError(
+            raise StaleClusterViewError(
                 f"no ControllerRevisions found for DaemonSet {meta.name(daemonset)}"
             )
         latest = max(revisions, key=lambda r: r.get("revision", 0))

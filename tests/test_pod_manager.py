@@ -59,8 +59,12 @@ class TestRevisionHash:
         assert mgr.get_daemonset_controller_revision_hash(ds) == "newhash"
 
     def test_daemonset_without_revisions_raises(self, client, provider):
+        from k8s_operator_libs_amd.upgrade.pod_manager import (
+            StaleClusterViewError,
+        )
+
         ds = DaemonSetBuilder("amdgpu-driver").build(client.cluster)
-        with pytest.raises(ValueError):
+        with pytest.raises(StaleClusterViewError):
             make_manager(client, provider).get_daemonset_controller_revision_hash(ds)
 
 
