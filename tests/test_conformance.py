@@ -451,8 +451,12 @@ class TestUpgradeLifecycleConformance:
         try:
             deadline = time.monotonic() + 60
             while time.monotonic() < deadline:
-                state = manager.build_state(ns, labels)
-                manager.apply_state(state, _policy())
+                try:
+                    state = manager.build_state(ns, labels)
+                    manager.apply_state(state, _policy())
+                except Exception:
+                    time.sleep(0.2)  # transient view; requeue like Reconcile
+                    continue
                 manager.wait_idle()
                 if _state_of(k8s, nodes[0]) == consts.UPGRADE_STATE_DONE:
                     break
@@ -503,8 +507,12 @@ class TestUpgradeLifecycleConformance:
         try:
             deadline = time.monotonic() + 60
             while time.monotonic() < deadline:
-                state = manager.build_state(ns, labels)
-                manager.apply_state(state, _policy())
+                try:
+                    state = manager.build_state(ns, labels)
+                    manager.apply_state(state, _policy())
+                except Exception:
+                    time.sleep(0.2)
+                    continue
                 manager.wait_idle()
                 if _state_of(k8s, nodes[0]) == consts.UPGRADE_STATE_DONE:
                     break
