@@ -146,7 +146,10 @@ class FakeCluster:
         self._lock = threading.RLock()
         # (apiVersion, kind) -> {(namespace, name): object}
         self._store: Dict[Tuple[str, str], Dict[Tuple[str, str], K8sObject]] = {}
-        self._rv_counter = 0
+        # starts at 1: a real apiserver never serves resourceVersion "0"
+        # (it is the magic watch value "any revision"), so neither may an
+        # empty cluster's LIST here — found by tests/test_property_watch.py
+        self._rv_counter = 1
         self._kinds = dict(_BUILTIN_KINDS)
         self._watches: Dict[Tuple[str, str], List[Watch]] = {}
         self._change_hooks: List[Callable[[str, K8sObject], None]] = []
