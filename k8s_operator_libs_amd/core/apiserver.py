@@ -150,6 +150,7 @@ def create_app(cluster: Optional[FakeCluster] = None):
                 namespace=namespace or None,
                 resource_version=resource_version,
                 label_selector=label_selector,
+                field_selector=qp.get("fieldSelector", ""),
             )
         except GoneError as exc:
             # expired before the stream even opened: stream a single ERROR
@@ -553,6 +554,7 @@ def _make_threaded_server(host, port, cluster):
                     api_version, kind, namespace=namespace or None,
                     resource_version=qp.get("resourceVersion") or None,
                     label_selector=qp.get("labelSelector", ""),
+                    field_selector=qp.get("fieldSelector", ""),
                 )
             except _Gone as exc:
                 # 200 + in-stream ERROR Status, the real apiserver shape

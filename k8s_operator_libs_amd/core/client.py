@@ -164,10 +164,11 @@ class FakeClient(Client):
         self.cluster.evict_pod(name, namespace)
 
     def watch(self, api_version, kind, namespace=None, resource_version=None,
-              label_selector=""):
+              label_selector="", field_selector=""):
         return self.cluster.watch(
             api_version, kind, namespace=namespace,
             resource_version=resource_version, label_selector=label_selector,
+            field_selector=field_selector,
         )
 
     def list_with_meta(self, api_version, kind, namespace=None,

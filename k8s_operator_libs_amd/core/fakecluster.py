@@ -69,33 +69,40 @@ class Watch:
         key: Tuple[str, str],
         namespace: Optional[str] = None,
         label_selector: str = "",
+        field_selector: str = "",
     ) -> None:
         self._cluster = cluster
         self._key = key
         self._namespace = namespace or None
-        self._selector = (
-            meta.parse_label_selector(label_selector) if label_selector else None
-        )
-        # object keys currently matching the selector (for the
-        # stops-matching -> DELETED transform); only used with a selector
+        selectors = []
+        if label_selector:
+            selectors.append(meta.parse_label_selector(label_selector))
+        if field_selector:
+            selectors.append(meta.parse_field_selector(field_selector))
+        self._selectors = selectors or None
+        # object keys currently matching the selectors (for the
+        # stops-matching -> DELETED transform); only used with selectors
         self._matched: set = set()
         self.events: "queue.Queue[Tuple[str, K8sObject]]" = queue.Queue()
         self._stopped = False
+
+    def _matches(self, snapshot: K8sObject) -> bool:
+        return all(s.matches_object(snapshot) for s in self._selectors)
 
     def _deliver(self, event_type: str, snapshot: K8sObject) -> None:
         """Apply namespace/selector filtering; called under the cluster lock
         so delivery order matches resourceVersion order."""
         if self._namespace is not None and meta.namespace(snapshot) != self._namespace:
             return
-        if self._selector is not None:
+        if self._selectors is not None:
             okey = (meta.namespace(snapshot), meta.name(snapshot))
-            matches = event_type != "DELETED" and self._selector.matches_object(snapshot)
+            matches = event_type != "DELETED" and self._matches(snapshot)
             was_matched = okey in self._matched
             if event_type == "DELETED":
                 # deliver if this watch saw the object OR the final snapshot
                 # matches (a resumed watch replaying a delete it never saw
                 # the ADDED for must still get the DELETED)
-                if not was_matched and not self._selector.matches_object(snapshot):
+                if not was_matched and not self._matches(snapshot):
                     return
                 self._matched.discard(okey)
             elif matches and not was_matched:
@@ -606,6 +613,7 @@ class FakeCluster:
         namespace: Optional[str] = None,
         resource_version: Optional[str] = None,
         label_selector: str = "",
+        field_selector: str = "",
     ) -> Watch:
         """Open a watch stream with Kubernetes resourceVersion semantics:
 
@@ -629,7 +637,9 @@ class FakeCluster:
             if key not in self._history:
                 self._history[key] = collections.deque(maxlen=self.WATCH_HISTORY)
                 self._history_start[key] = self._rv_counter
-            w = Watch(self, key, namespace=namespace, label_selector=label_selector)
+            w = Watch(self, key, namespace=namespace,
+                      label_selector=label_selector,
+                      field_selector=field_selector)
             if resource_version == "0":
                 for obj in self._store.get(key, {}).values():
                     w._deliver("ADDED", meta.deep_copy(obj))

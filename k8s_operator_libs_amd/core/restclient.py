@@ -366,7 +366,7 @@ class RestClient(Client):
         self._raise_for(resp)
 
     def watch(self, api_version: str, kind: str, namespace=None,
-              resource_version=None, label_selector=""):
+              resource_version=None, label_selector="", field_selector=""):
         """Open a Kubernetes watch stream (``?watch=true``) with standard
         query parameters: ``resourceVersion`` anchoring, ``labelSelector``
         filtering, namespace-scoped paths, and ``allowWatchBookmarks`` so
@@ -378,7 +378,8 @@ class RestClient(Client):
         event, both of which the informer answers with relist."""
         return _HttpWatch(self, api_version, kind, namespace=namespace,
                           resource_version=resource_version,
-                          label_selector=label_selector)
+                          label_selector=label_selector,
+                          field_selector=field_selector)
 
     # -- discovery (for crdutil.wait_for_crds) ----------------------------------
 
@@ -428,7 +429,7 @@ class _HttpWatch:
 
     def __init__(self, client: RestClient, api_version: str, kind: str,
                  namespace=None, resource_version=None,
-                 label_selector="") -> None:
+                 label_selector="", field_selector="") -> None:
         import queue
         import threading
 
@@ -441,6 +442,8 @@ class _HttpWatch:
             params["resourceVersion"] = str(resource_version)
         if label_selector:
             params["labelSelector"] = label_selector
+        if field_selector:
+            params["fieldSelector"] = field_selector
         self._error: Optional[BaseException] = None
 
         def reader():

@@ -351,3 +351,28 @@ class TestWireProtocol:
         assert got["metadata"]["name"] == "pdb1"
         assert rest._kinds[("policy/v1", "PodDisruptionBudget")] == (
             "poddisruptionbudgets", True)
+
+
+def test_field_selector_watch():
+    """Field-selector watches (spec.nodeName) with the same
+    transform semantics as label selectors."""
+    c = FakeCluster()
+    w = c.watch("v1", "Pod", field_selector="spec.nodeName=n1")
+    c.create(pod("p1"))          # nodeName n1 (pod() default)
+    ev = w.next(0.5)
+    assert ev[0] == "ADDED" and ev[1]["metadata"]["name"] == "p1"
+    c.create({"apiVersion": "v1", "kind": "Pod",
+              "metadata": {"name": "p2", "namespace": "default"},
+              "spec": {"nodeName": "other"}})
+    assert w.next(0.1) is None  # other node filtered out
+    w.stop()
+
+
+def test_field_selector_watch_over_http(rest, server):
+    w = rest.watch("v1", "Pod", field_selector="spec.nodeName=n1")
+    try:
+        rest.create(pod("f1"))
+        ev = w.next(5.0)
+        assert ev[0] == "ADDED" and ev[1]["metadata"]["name"] == "f1"
+    finally:
+        w.stop()
