@@ -108,3 +108,91 @@ def test_converges_despite_random_faults(seed):
     for p in client.list_pods(namespace="amd-gpu-operator",
                               label_selector="app=amdgpu-driver-daemonset"):
         assert p["metadata"]["labels"]["controller-revision-hash"] == "new"
+
+
+@settings(max_examples=4 * _SCALE, deadline=None)
+@given(seed=st.integers(min_value=0, max_value=10_000))
+def test_converges_despite_faults_over_the_wire(seed):
+    """The same chaos campaign through the FULL production stack — state
+    machine -> CachedClient informers -> REST -> HTTP apiserver — with an
+    extra wire-only fault: the apiserver itself restarting mid-campaign.
+    Transient staleness (BuildStateError / NotFound / missing revisions) is
+    requeued exactly like a Reconcile error."""
+    import socket
+    import time as _time
+
+    from k8s_operator_libs_amd.core.apiserver import start_apiserver
+    from k8s_operator_libs_amd.core.cache import CachedClient
+    from k8s_operator_libs_amd.core.errors import ApiError, NotFoundError
+    from k8s_operator_libs_amd.core.fakecluster import FakeCluster
+    from k8s_operator_libs_amd.core.restclient import RestClient
+    from k8s_operator_libs_amd.upgrade.pod_manager import StaleClusterViewError
+    from k8s_operator_libs_amd.upgrade.state_manager import BuildStateError
+
+    rng = random.Random(seed)
+    n_nodes = rng.randrange(2, 5)
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cluster = FakeCluster()
+    handle = start_apiserver(port=port, cluster=cluster)
+    rest = RestClient(handle.url)
+    cached = CachedClient(rest)
+
+    class W:
+        pass
+
+    W.cluster = cluster
+    ds, _ = setup_cluster(W, n_nodes=n_nodes, pod_hash="old", ds_hash="new")
+    SimDaemonSetController(cluster, ds, current_hash="new")
+    sim_client = FakeClient(cluster)  # faults act directly on the store
+    manager = ClusterUpgradeStateManager(cached)
+    pol = policy(maxParallelUpgrades=rng.randrange(0, 3),
+                 maxUnavailable="100%",
+                 drainSpec={"enable": bool(rng.randrange(2))})
+    state_key = util.get_upgrade_state_label_key()
+
+    def tick():
+        try:
+            manager.reconcile("amd-gpu-operator",
+                              {"app": "amdgpu-driver-daemonset"}, pol,
+                              converge=True)
+        except (BuildStateError, NotFoundError, StaleClusterViewError,
+                ApiError):
+            _time.sleep(0.02)
+
+    try:
+        restarts = 0
+        for _ in range(rng.randrange(4, 10)):
+            tick()
+            if restarts < 2 and rng.randrange(3) == 0:
+                handle.stop()
+                _time.sleep(0.05)
+                handle = start_apiserver(port=port, cluster=cluster)
+                restarts += 1
+            for _ in range(rng.randrange(0, 2)):
+                _inject_fault(rng, sim_client, n_nodes)
+
+        for i in range(n_nodes):
+            cluster.patch("v1", "Node", f"node-{i}",
+                          {"status": {"conditions": [{"type": "Ready",
+                                                      "status": "True"}]}})
+        deadline = _time.monotonic() + 60
+        while _time.monotonic() < deadline:
+            tick()
+            states = [n["metadata"].get("labels", {}).get(state_key, "")
+                      for n in cluster.list("v1", "Node")]
+            if all(s == consts.UPGRADE_STATE_DONE for s in states):
+                break
+            _time.sleep(0.02)
+        assert all(
+            n["metadata"].get("labels", {}).get(state_key)
+            == consts.UPGRADE_STATE_DONE
+            for n in cluster.list("v1", "Node")
+        ), f"seed {seed} did not converge over the wire: {states}"
+    finally:
+        manager.common.drain_manager.wait_idle()
+        manager.common.pod_manager.wait_idle()
+        cached.stop()
+        rest.close()
+        handle.stop()

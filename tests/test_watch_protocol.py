@@ -376,3 +376,49 @@ def test_field_selector_watch_over_http(rest, server):
         assert ev[0] == "ADDED" and ev[1]["metadata"]["name"] == "f1"
     finally:
         w.stop()
+
+
+class TestUvicornEngineParity:
+    """The FastAPI/uvicorn engine stays wire-compatible with the default
+    threaded engine (both serve the same protocol; the threaded one is the
+    benchmark substrate)."""
+
+    @pytest.fixture(scope="class")
+    def uv(self):
+        handle = start_apiserver(engine="uvicorn")
+        client = RestClient(handle.url)
+        yield client, handle
+        client.close()
+        handle.stop()
+
+    def test_crud_and_rv_watch(self, uv):
+        rest, server = uv
+        rest.create(node("uv1"))
+        items, rv = rest.list_with_meta("v1", "Node")
+        assert rv and len(items) == 1
+        w = rest.watch("v1", "Node", resource_version=rv)
+        try:
+            rest.patch("v1", "Node", "uv1", {"metadata": {"labels": {"u": "1"}}})
+            ev = w.next(5.0)
+            while ev and ev[0] in ("BOOKMARK",):
+                ev = w.next(5.0)
+            assert ev[0] == "MODIFIED"
+        finally:
+            w.stop()
+
+    def test_pagination_and_eviction(self, uv):
+        rest, server = uv
+        for i in range(3):
+            rest.create({"apiVersion": "v1", "kind": "Pod",
+                         "metadata": {"name": f"uvp-{i}", "namespace": "default"},
+                         "spec": {"containers": [{"name": "c", "image": "x"}]}})
+        old = rest.LIST_PAGE_SIZE
+        try:
+            rest.LIST_PAGE_SIZE = 2
+            items, _ = rest.list_with_meta("v1", "Pod", namespace="default")
+            assert len(items) == 3
+        finally:
+            rest.LIST_PAGE_SIZE = old
+        rest.evict_pod("uvp-0", "default")
+        with pytest.raises(Exception):
+            rest.get("v1", "Pod", "uvp-0", "default")
