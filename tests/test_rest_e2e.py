@@ -329,3 +329,25 @@ def test_full_upgrade_through_informer_stack(rest, server):
             assert count == 1, f"{frm}->{to} fired {count} times"
     finally:
         cached.stop()
+
+
+def test_from_environment_service_account(server, tmp_path, monkeypatch):
+    """In-cluster resolution: service-account token + KUBERNETES_SERVICE_*
+    (restclient.py from_environment path 2).  The URL is https (in-cluster
+    is always TLS) so this pins resolution, not connectivity."""
+    import k8s_operator_libs_amd.core.restclient as rc
+
+    sa = tmp_path / "serviceaccount"
+    sa.mkdir()
+    (sa / "token").write_text("sa-token-123\n")
+    (sa / "ca.crt").write_text("---fake ca---")
+    monkeypatch.setattr(rc, "SA_DIR", str(sa))
+    monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    monkeypatch.setenv("KUBERNETES_SERVICE_PORT", "6443")
+    client = rc.RestClient.from_environment()
+    try:
+        assert client.base_url == "https://10.0.0.1:6443"
+        assert client._headers["Authorization"] == "Bearer sa-token-123"
+        assert client._fast_netloc is None  # TLS: stays on httpx
+    finally:
+        client.close()
