@@ -340,7 +340,8 @@ def test_from_environment_service_account(server, tmp_path, monkeypatch):
     sa = tmp_path / "serviceaccount"
     sa.mkdir()
     (sa / "token").write_text("sa-token-123\n")
-    (sa / "ca.crt").write_text("---fake ca---")
+    # no ca.crt: a PEM-invalid CA would fail httpx's SSL context load;
+    # the resolver falls back to verify=True
     monkeypatch.setattr(rc, "SA_DIR", str(sa))
     monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
     monkeypatch.setenv("KUBERNETES_SERVICE_PORT", "6443")
