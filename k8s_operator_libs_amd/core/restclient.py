@@ -453,6 +453,7 @@ class _HttpWatch:
                 with client._http.stream(
                     "GET", path, params=params, timeout=None,
                 ) as resp:
+                    self._resp = resp  # stop() closes it to unblock instantly
                     if resp.status_code == 410:
                         from .errors import GoneError
 
@@ -488,6 +489,7 @@ class _HttpWatch:
             finally:
                 self._connected.set()  # never leave a waiter hanging
 
+        self._resp = None
         self._thread = threading.Thread(target=reader, daemon=True)
         self._thread.start()
         self._connected.wait(10.0)
@@ -504,6 +506,14 @@ class _HttpWatch:
 
     def stop(self) -> None:
         self._stop.set()
+        resp = self._resp
+        if resp is not None:
+            # close the stream so the blocked reader exits immediately
+            # instead of waiting for the next keep-alive frame
+            try:
+                resp.close()
+            except Exception:
+                pass
 
     def alive(self) -> bool:
         """False once the HTTP stream has ended (server drop or stop)."""
