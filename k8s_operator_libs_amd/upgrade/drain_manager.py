@@ -37,6 +37,14 @@ class DrainConfiguration:
 
 
 class DrainManager:
+    #: Bounded worker concurrency.  The reference spawns one goroutine per
+    #: node (drain_manager.go:98-137) and lets the Go scheduler absorb any
+    #: width; Python threads + the GIL do not — a 64-node drain wave at
+    #: maxParallelUpgrades=32 measured SLOWER than at 8 from thread thrash
+    #: and apiserver contention (BASELINE.md window sweep).  Excess workers
+    #: queue on a semaphore; dedup/ordering semantics are unchanged.
+    MAX_CONCURRENT_NODE_WORKERS = 8
+
     def __init__(
         self,
         client: Client,
@@ -49,6 +57,7 @@ class DrainManager:
         self._draining_nodes = util.StringSet()
         self._workers: List[threading.Thread] = []
         self._workers_lock = threading.Lock()
+        self._worker_slots = threading.Semaphore(self.MAX_CONCURRENT_NODE_WORKERS)
 
     def schedule_nodes_drain(self, config: DrainConfiguration) -> None:
         if config.spec is None:
@@ -74,6 +83,11 @@ class DrainManager:
 
     def _drain_worker(self, node: K8sObject, spec: DrainSpec) -> None:
         node_name = meta.name(node)
+        with self._worker_slots:
+            self._drain_worker_inner(node, spec, node_name)
+
+    def _drain_worker_inner(self, node: K8sObject, spec: DrainSpec,
+                            node_name: str) -> None:
         try:
             try:
                 # Cordon first: drains only make sense on unschedulable nodes

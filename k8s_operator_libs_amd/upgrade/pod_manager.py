@@ -83,6 +83,8 @@ class PodManager:
         self._nodes_in_progress = util.StringSet()
         self._workers: List[threading.Thread] = []
         self._workers_lock = threading.Lock()
+        # bounded eviction-worker concurrency (see DrainManager rationale)
+        self._worker_slots = threading.Semaphore(self.MAX_CONCURRENT_NODE_WORKERS)
         # (namespace, name, resourceVersion) -> latest revision hash; a DS's
         # hash can only change when the DS object itself changes, so this
         # collapses the per-node ControllerRevision LISTs of a reconcile pass
@@ -288,8 +290,14 @@ class PodManager:
         """(pod_manager.go:371-391)"""
         return pod.get("status", {}).get("phase") in ("Running", "Pending")
 
+    MAX_CONCURRENT_NODE_WORKERS = 8
+
     def _spawn(self, target, *args) -> None:
-        t = threading.Thread(target=target, args=args, daemon=True)
+        def bounded():
+            with self._worker_slots:
+                target(*args)
+
+        t = threading.Thread(target=bounded, daemon=True)
         with self._workers_lock:
             self._workers = [w for w in self._workers if w.is_alive()]
             self._workers.append(t)
