@@ -59,6 +59,7 @@ def serve_metrics(manager, port: int):
         def log_message(self, *args):
             pass
 
+    HTTPServer.allow_reuse_address = True  # survive TIME_WAIT on restart
     server = HTTPServer(("127.0.0.1", port), Handler)
     threading.Thread(target=server.serve_forever, daemon=True).start()
     return server

@@ -52,7 +52,7 @@ def _substrates():
     subs = ["mini"]
     if find_assets() is not None:
         subs.append("envtest")
-    if os.environ.get("CONFORMANCE_URL"):
+    if os.environ.get("CONFORMANCE_URL") or os.environ.get("CONFORMANCE_KUBECONFIG"):
         subs.append("external")
     return subs
 
@@ -77,10 +77,14 @@ def substrate(request):
         client.close()
         cluster.stop()
     else:
-        url = os.environ["CONFORMANCE_URL"]
-        client = RestClient(url, token=os.environ.get("CONFORMANCE_TOKEN"),
-                            verify=False)
-        yield client, url, True
+        kubeconfig = os.environ.get("CONFORMANCE_KUBECONFIG")
+        if kubeconfig:
+            client = RestClient._from_kubeconfig(kubeconfig)
+        else:
+            client = RestClient(os.environ["CONFORMANCE_URL"],
+                                token=os.environ.get("CONFORMANCE_TOKEN"),
+                                verify=False)
+        yield client, client.base_url, True
         client.close()
 
 
