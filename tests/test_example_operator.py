@@ -101,7 +101,6 @@ def test_production_entrypoint_with_leader_election(monkeypatch):
 
     from k8s_operator_libs_amd.core.apiserver import start_apiserver
     from k8s_operator_libs_amd.testing import (
-        DRIVER_LABELS,
         DRIVER_NS,
         SimDaemonSetController,
     )
