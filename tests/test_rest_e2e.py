@@ -352,3 +352,94 @@ def test_from_environment_service_account(server, tmp_path, monkeypatch):
         assert client._fast_netloc is None  # TLS: stays on httpx
     finally:
         client.close()
+
+
+def test_transient_5xx_retried(monkeypatch):
+    """client-go-style retry: 503s from a briefly-unhealthy apiserver are
+    retried with backoff; the request ultimately succeeds (restclient.py
+    _request).  Pinned with a raw socket server so both transports (fast
+    http.client and httpx) see identical wire bytes."""
+    import http.server
+    import json as _json
+    import socketserver
+    import threading
+
+    state = {"fails": 2, "requests": 0}
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"
+
+        def log_message(self, *a):
+            pass
+
+        def do_GET(self):
+            state["requests"] += 1
+            if state["fails"] > 0:
+                state["fails"] -= 1
+                body = b'{"kind":"Status","message":"apiserver warming up"}'
+                self.send_response(503)
+            else:
+                body = _json.dumps({"apiVersion": "v1", "kind": "Node",
+                                    "metadata": {"name": "n1"}}).encode()
+                self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+    class Srv(socketserver.ThreadingMixIn, http.server.HTTPServer):
+        daemon_threads = True
+        allow_reuse_address = True
+
+    srv = Srv(("127.0.0.1", 0), Handler)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        client = RestClient(f"http://127.0.0.1:{srv.server_address[1]}")
+        node = client.get_node("n1")
+        assert node["metadata"]["name"] == "n1"
+        assert state["requests"] == 3  # 2 failures + 1 success
+        client.close()
+    finally:
+        srv.shutdown()
+        srv.server_close()
+
+
+def test_retries_exhausted_surface_api_error(monkeypatch):
+    import http.server
+    import socketserver
+    import threading
+
+    import pytest as _pytest
+
+    from k8s_operator_libs_amd.core.errors import ApiError
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"
+
+        def log_message(self, *a):
+            pass
+
+        def do_GET(self):
+            body = b'{"kind":"Status","message":"still down"}'
+            self.send_response(503)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+    class Srv(socketserver.ThreadingMixIn, http.server.HTTPServer):
+        daemon_threads = True
+        allow_reuse_address = True
+
+    srv = Srv(("127.0.0.1", 0), Handler)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        client = RestClient(f"http://127.0.0.1:{srv.server_address[1]}",
+                            retries=1)
+        with _pytest.raises(ApiError) as exc:
+            client.get_node("n1")
+        assert exc.value.code == 503
+        client.close()
+    finally:
+        srv.shutdown()
+        srv.server_close()
