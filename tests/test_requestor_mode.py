@@ -385,3 +385,75 @@ class TestSharedWindowEndToEnd:
             assert pods and all(
                 p["metadata"]["labels"]["controller-revision-hash"] == "new"
                 for p in pods)
+
+
+class TestSharedWindowOverWire:
+    def test_shared_maintenance_window_over_http(self):
+        """The shared-requestor optimistic-lock protocol at the wire level:
+        two operators (amdgpu + anic) through separate RestClients against
+        the HTTP apiserver share one NodeMaintenance per node; the
+        resourceVersion-locked joins/releases travel real HTTP (409s
+        included).  In-process variant: TestSharedWindowEndToEnd."""
+        from k8s_operator_libs_amd.core.apiserver import start_apiserver
+        from k8s_operator_libs_amd.core.restclient import RestClient
+        from builders import DaemonSetBuilder, driver_pod_for, make_controller_revision
+        from simenv import SimMaintenanceOperator
+
+        handle = start_apiserver()
+        gpu_rest = RestClient(handle.url)
+        nic_rest = RestClient(handle.url)
+        try:
+            nm_name = f"{DEFAULT_NODE_MAINTENANCE_NAME_PREFIX}-node-0"
+            SimMaintenanceOperator(handle.cluster)
+
+            class W:
+                cluster = handle.cluster
+
+            util.set_driver_name("amdgpu")
+            gpu_ds, _ = setup_cluster(W, pod_hash="old", ds_hash="new",
+                                      ds_name="amdgpu-driver")
+            SimDaemonSetController(handle.cluster, gpu_ds, current_hash="new")
+            gpu_mgr = make_manager(gpu_rest, requestor_id="amd.gpu.operator")
+
+            util.set_driver_name("anic")
+            nic_labels = {"app": "anic-driver-daemonset"}
+            nic_ds = DaemonSetBuilder("anic-driver", labels=nic_labels) \
+                .with_desired_number_scheduled(1).build(handle.cluster)
+            make_controller_revision(nic_ds, "new", revision=2,
+                                     cluster=handle.cluster)
+            make_controller_revision(nic_ds, "old", revision=1,
+                                     cluster=handle.cluster)
+            driver_pod_for(nic_ds, "node-0", hash_="old").build(handle.cluster)
+            SimDaemonSetController(handle.cluster, nic_ds, current_hash="new")
+            nic_mgr = make_manager(nic_rest, requestor_id="amd.network.operator")
+
+            pol = policy(drainSpec={"enable": True})
+            saw_shared = False
+            gpu_done = nic_done = False
+            for _ in range(25):
+                util.set_driver_name("amdgpu")
+                gpu_mgr.reconcile(DRIVER_NS, DRIVER_LABELS, pol)
+                util.set_driver_name("anic")
+                nic_mgr.reconcile(DRIVER_NS, nic_labels, pol)
+                try:
+                    nm = gpu_rest.get(NM_API, "NodeMaintenance", nm_name,
+                                      "default")
+                    if nm["spec"].get("additionalRequestors"):
+                        saw_shared = True
+                except NotFoundError:
+                    pass
+                util.set_driver_name("amdgpu")
+                gpu_done = state_of(gpu_rest, "node-0") == consts.UPGRADE_STATE_DONE
+                util.set_driver_name("anic")
+                nic_done = state_of(nic_rest, "node-0") == consts.UPGRADE_STATE_DONE
+                if gpu_done and nic_done:
+                    break
+            assert gpu_done and nic_done, (gpu_done, nic_done)
+            assert saw_shared, "operators never shared the window over HTTP"
+            with pytest.raises(NotFoundError):
+                gpu_rest.get(NM_API, "NodeMaintenance", nm_name, "default")
+            assert not gpu_rest.get_node("node-0")["spec"].get("unschedulable")
+        finally:
+            gpu_rest.close()
+            nic_rest.close()
+            handle.stop()
