@@ -31,6 +31,14 @@ class BadRequestError(ApiError):
     code = 400
 
 
+class InvalidError(ApiError):
+    """Schema-invalid object (HTTP 422, status reason ``Invalid``) — what a
+    real apiserver returns when a custom resource violates its CRD's
+    structural openAPIV3Schema."""
+
+    code = 422
+
+
 class GoneError(ApiError):
     """Watch resume window expired (HTTP 410, status reason ``Expired``) —
     the client must relist and re-watch from the fresh resourceVersion."""

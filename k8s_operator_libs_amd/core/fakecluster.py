@@ -167,6 +167,8 @@ class FakeCluster:
         self._history: Dict[Tuple[str, str], "collections.deque"] = {}
         # RV from which each kind's history is complete
         self._history_start: Dict[Tuple[str, str], int] = {}
+        # structural schemas of CRD-registered kinds, enforced on CR writes
+        self._cr_schemas: Dict[Tuple[str, str], dict] = {}
         # spec.nodeName index for pods: nodeName -> {(ns, name)}; keeps
         # per-node pod LISTs (the reconcile loop's hottest query) O(pods on
         # node) instead of O(all pods)
@@ -179,7 +181,9 @@ class FakeCluster:
             self._kinds[(api_version, kind)] = (plural, namespaced)
 
     def register_crd(self, crd: K8sObject) -> None:
-        """Make a created CustomResourceDefinition's kind servable."""
+        """Make a created CustomResourceDefinition's kind servable and
+        record its structural schema — subsequent CRs of the kind are
+        validated against it (422 Invalid), like a real apiserver."""
         spec = crd.get("spec", {})
         group = spec.get("group", "")
         names = spec.get("names", {})
@@ -188,7 +192,11 @@ class FakeCluster:
         namespaced = spec.get("scope", "Namespaced") == "Namespaced"
         for ver in spec.get("versions", []):
             if ver.get("served", True):
-                self.register_kind(f"{group}/{ver['name']}", kind, plural, namespaced)
+                api_version = f"{group}/{ver['name']}"
+                self.register_kind(api_version, kind, plural, namespaced)
+                schema = (ver.get("schema") or {}).get("openAPIV3Schema")
+                if schema:
+                    self._cr_schemas[(api_version, kind)] = schema
 
     def lookup_kind(self, api_version: str, kind: str) -> Tuple[str, bool]:
         try:
@@ -205,6 +213,20 @@ class FakeCluster:
         return None
 
     # -- internal helpers ----------------------------------------------------
+
+    def _validate_cr(self, api_version: str, kind: str, obj: K8sObject) -> None:
+        schema = self._cr_schemas.get((api_version, kind))
+        if schema is None:
+            return
+        from .errors import InvalidError
+
+        body = {k: v for k, v in obj.items()
+                if k not in ("apiVersion", "kind", "metadata")}
+        errs = meta.validate_structural_schema(body, schema, path=kind)
+        if errs:
+            raise InvalidError(
+                f"{kind} is invalid: " + "; ".join(errs[:5])
+            )
 
     def _next_rv(self) -> str:
         with self._lock:
@@ -298,6 +320,7 @@ class FakeCluster:
             bucket = self._bucket(api_version, kind)
             if (ns, name_) in bucket:
                 raise AlreadyExistsError(f"{kind} {ns}/{name_} already exists")
+            self._validate_cr(api_version, kind, obj)
             # respect a caller-provided uid (snapshot load, fixtures with
             # pre-wired ownerReferences); assign one otherwise
             md["uid"] = md.get("uid") or str(uuid.uuid4())
@@ -437,6 +460,7 @@ class FakeCluster:
                 raise ConflictError(
                     f"{kind} {ns}/{name_}: resourceVersion {rv} is stale"
                 )
+            self._validate_cr(api_version, kind, obj)
             # Immutable server-side fields carry over.
             obj["metadata"]["uid"] = stored["metadata"]["uid"]
             obj["metadata"]["creationTimestamp"] = stored["metadata"]["creationTimestamp"]
@@ -482,6 +506,10 @@ class FakeCluster:
                     f"{kind} {ns}/{name}: resourceVersion {patch_rv} is stale"
                 )
             old_node = stored.get("spec", {}).get("nodeName", "") if kind == "Pod" else None
+            if (api_version, kind) in self._cr_schemas:
+                preview = meta.deep_copy(stored)
+                meta.json_merge_patch(preview, patch)
+                self._validate_cr(api_version, kind, preview)
             meta.json_merge_patch(stored, patch)
             stored["metadata"]["name"] = name  # patches cannot rename
             stored["metadata"]["resourceVersion"] = self._next_rv()

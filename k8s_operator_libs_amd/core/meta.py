@@ -265,3 +265,66 @@ def json_merge_patch(target: Any, patch: Any) -> Any:
         else:
             target[key] = deep_copy(value)
     return target
+
+
+def validate_structural_schema(obj, schema, path="spec"):
+    """Validate ``obj`` against a structural-openAPIV3Schema subset — the
+    checks a real apiserver applies to custom resources (types, required,
+    properties, additionalProperties=False, enum,
+    x-kubernetes-int-or-string).  Returns a list of violation strings
+    (empty = valid); callers map non-empty to HTTP 422 Invalid."""
+    errs = []
+    if schema is None:
+        return errs
+    if schema.get("x-kubernetes-int-or-string"):
+        if not isinstance(obj, (int, str)) or isinstance(obj, bool):
+            errs.append(f"{path}: expected integer-or-string")
+        return errs
+    if schema.get("x-kubernetes-preserve-unknown-fields"):
+        return errs
+    stype = schema.get("type")
+    if stype == "object" or (stype is None and "properties" in schema):
+        if not isinstance(obj, dict):
+            errs.append(f"{path}: expected object")
+            return errs
+        props = schema.get("properties", {})
+        for req in schema.get("required", []):
+            if req not in obj:
+                errs.append(f"{path}.{req}: required field missing")
+        addl = schema.get("additionalProperties", True)
+        for key, value in obj.items():
+            if key in props:
+                errs.extend(
+                    validate_structural_schema(value, props[key], f"{path}.{key}")
+                )
+            elif addl is False:
+                errs.append(f"{path}.{key}: unknown field")
+            elif isinstance(addl, dict):
+                errs.extend(
+                    validate_structural_schema(value, addl, f"{path}.{key}")
+                )
+    elif stype == "array":
+        if not isinstance(obj, list):
+            errs.append(f"{path}: expected array")
+            return errs
+        items = schema.get("items")
+        if items:
+            for i, v in enumerate(obj):
+                errs.extend(
+                    validate_structural_schema(v, items, f"{path}[{i}]")
+                )
+    elif stype == "string":
+        if not isinstance(obj, str):
+            errs.append(f"{path}: expected string")
+    elif stype == "integer":
+        if not isinstance(obj, int) or isinstance(obj, bool):
+            errs.append(f"{path}: expected integer")
+    elif stype == "number":
+        if not isinstance(obj, (int, float)) or isinstance(obj, bool):
+            errs.append(f"{path}: expected number")
+    elif stype == "boolean":
+        if not isinstance(obj, bool):
+            errs.append(f"{path}: expected boolean")
+    if "enum" in schema and obj not in schema["enum"]:
+        errs.append(f"{path}: {obj!r} not in enum {schema['enum']}")
+    return errs

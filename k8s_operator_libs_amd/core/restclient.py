@@ -264,6 +264,10 @@ class RestClient(Client):
             raise ConflictError(message)
         if resp.status_code == 400:
             raise BadRequestError(message)
+        if resp.status_code == 422:
+            from .errors import InvalidError
+
+            raise InvalidError(message)
         err = ApiError(message)
         err.code = resp.status_code
         raise err

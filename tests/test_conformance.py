@@ -524,3 +524,73 @@ class TestUpgradeLifecycleConformance:
             assert _state_of(k8s, nodes[0]) == consts.UPGRADE_STATE_DONE
         finally:
             adapter.stop()
+
+
+class TestCrValidationConformance:
+    def test_invalid_cr_rejected_with_422(self, k8s, tmp_path):
+        """Real apiservers validate CRs against the CRD's structural
+        schema (422 Invalid); the mini-apiserver enforces the same."""
+        import yaml
+
+        from k8s_operator_libs_amd.core.errors import InvalidError
+        from k8s_operator_libs_amd.crdutil import CRD_OPERATION_APPLY, process_crds
+
+        group = f"val{uuid.uuid4().hex[:6]}.amd.com"
+        crd = {
+            "apiVersion": "apiextensions.k8s.io/v1",
+            "kind": "CustomResourceDefinition",
+            "metadata": {"name": f"gauges.{group}"},
+            "spec": {
+                "group": group, "scope": "Namespaced",
+                "names": {"kind": "Gauge", "plural": "gauges",
+                          "singular": "gauge"},
+                "versions": [{
+                    "name": "v1", "served": True, "storage": True,
+                    "schema": {"openAPIV3Schema": {
+                        "type": "object",
+                        "properties": {"spec": {
+                            "type": "object",
+                            "required": ["size"],
+                            "properties": {
+                                "size": {"type": "integer"},
+                                "mode": {"type": "string",
+                                         "enum": ["fast", "safe"]},
+                            },
+                            "additionalProperties": False,
+                        }},
+                    }},
+                }],
+            },
+        }
+        path = tmp_path / "gauge-crd.yaml"
+        path.write_text(yaml.safe_dump(crd))
+        process_crds(k8s, [str(path)], CRD_OPERATION_APPLY)
+        k8s.register_kind(f"{group}/v1", "Gauge", "gauges", True)
+
+        ok = k8s.create({"apiVersion": f"{group}/v1", "kind": "Gauge",
+                         "metadata": {"name": "g1", "namespace": "default"},
+                         "spec": {"size": 3, "mode": "fast"}})
+        assert ok["spec"]["size"] == 3
+        # wrong type
+        with pytest.raises(InvalidError):
+            k8s.create({"apiVersion": f"{group}/v1", "kind": "Gauge",
+                        "metadata": {"name": "g2", "namespace": "default"},
+                        "spec": {"size": "three"}})
+        # missing required
+        with pytest.raises(InvalidError):
+            k8s.create({"apiVersion": f"{group}/v1", "kind": "Gauge",
+                        "metadata": {"name": "g3", "namespace": "default"},
+                        "spec": {"mode": "fast"}})
+        # unknown field under additionalProperties: false
+        with pytest.raises(InvalidError):
+            k8s.create({"apiVersion": f"{group}/v1", "kind": "Gauge",
+                        "metadata": {"name": "g4", "namespace": "default"},
+                        "spec": {"size": 1, "bogus": True}})
+        # enum violation via PATCH of an existing object
+        with pytest.raises(InvalidError):
+            k8s.patch(f"{group}/v1", "Gauge", "g1",
+                      {"spec": {"mode": "yolo"}}, "default")
+        # valid patch still works
+        got = k8s.patch(f"{group}/v1", "Gauge", "g1",
+                        {"spec": {"size": 9}}, "default")
+        assert got["spec"]["size"] == 9
