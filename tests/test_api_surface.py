@@ -180,3 +180,33 @@ def test_controller_exported_from_upgrade_package():
     from k8s_operator_libs_amd import upgrade
 
     assert upgrade.UpgradeController
+
+
+def test_round2_surfaces():
+    """Round-2 additions stay exported: envtest launcher, conformance sim
+    adapters, watch-protocol APIs, linter, leader-election controller."""
+    from k8s_operator_libs_amd.core.cache import CachedClient
+    from k8s_operator_libs_amd.core.client import Client
+    from k8s_operator_libs_amd.core.errors import GoneError, InvalidError
+    from k8s_operator_libs_amd.core.fakecluster import FakeCluster
+    from k8s_operator_libs_amd.testing import ClientHookAdapter, SimKubelet
+    from k8s_operator_libs_amd.testing.envtest import (
+        EnvtestCluster,
+        find_assets,
+        start_envtest,
+    )
+    from k8s_operator_libs_amd.upgrade.controller import UpgradeController
+    from k8s_operator_libs_amd.upgrade.pod_manager import StaleClusterViewError
+
+    # RV-anchored watch + pagination + RV barrier APIs
+    assert callable(FakeCluster.list_with_meta)
+    assert callable(FakeCluster.list_paged)
+    assert callable(FakeCluster.current_rv)
+    assert callable(CachedClient.wait_for_resource_version)
+    assert callable(Client.list_with_meta)
+    assert callable(UpgradeController.run_with_leader_election)
+    assert GoneError.code == 410 and InvalidError.code == 422
+    assert issubclass(StaleClusterViewError, RuntimeError)
+    assert callable(find_assets) and callable(start_envtest)
+    assert hasattr(EnvtestCluster, "READY_TIMEOUT")
+    assert callable(ClientHookAdapter) and callable(SimKubelet)
