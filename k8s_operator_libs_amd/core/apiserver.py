@@ -290,8 +290,9 @@ def create_app(cluster: Optional[FakeCluster] = None):
 
         def run():
             api_version, kind = _resolve(group, version, plural)
-            status_patch = {"status": body.get("status", body)}
-            return cluster.patch(api_version, kind, name, status_patch, namespace)
+            return cluster.patch_status(
+                api_version, kind, name, body.get("status", body), namespace
+            )
         return _handle(run)
 
     app.add_api_route("/api/{version}/{plural}/{name}/status",
@@ -628,9 +629,9 @@ def _make_threaded_server(host, port, cluster):
                 kind = self._resolve(api_version, plural)
                 body = self._read_body()
                 if sub == "status":
-                    self._send_json(cluster.patch(
+                    self._send_json(cluster.patch_status(
                         api_version, kind, name,
-                        {"status": body.get("status", body)}, namespace))
+                        body.get("status", body), namespace))
                     return
                 if namespace:
                     body.setdefault("metadata", {})["namespace"] = namespace
@@ -643,7 +644,10 @@ def _make_threaded_server(host, port, cluster):
                 kind = self._resolve(api_version, plural)
                 body = self._read_body()
                 if sub == "status":
-                    body = {"status": body.get("status", body)}
+                    self._send_json(cluster.patch_status(
+                        api_version, kind, name,
+                        body.get("status", body), namespace))
+                    return
                 self._send_json(cluster.patch(api_version, kind, name, body,
                                               namespace))
             self._handle(run)

@@ -312,6 +312,10 @@ class CachedClient(Client):
     def evict_pod(self, name, namespace):
         self._delegate.evict_pod(name, namespace)
 
+    def patch_status(self, api_version, kind, name, status, namespace=""):
+        return self._delegate.patch_status(api_version, kind, name, status,
+                                           namespace)
+
     def watch(self, api_version, kind, namespace=None, resource_version=None,
               label_selector=""):
         return self._delegate.watch(

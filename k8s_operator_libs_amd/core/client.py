@@ -163,6 +163,10 @@ class FakeClient(Client):
     def evict_pod(self, name, namespace):
         self.cluster.evict_pod(name, namespace)
 
+    def patch_status(self, api_version, kind, name, status, namespace=""):
+        return self.cluster.patch_status(api_version, kind, name, status,
+                                         namespace)
+
     def watch(self, api_version, kind, namespace=None, resource_version=None,
               label_selector="", field_selector=""):
         return self.cluster.watch(

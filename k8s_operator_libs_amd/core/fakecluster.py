@@ -169,6 +169,11 @@ class FakeCluster:
         self._history_start: Dict[Tuple[str, str], int] = {}
         # structural schemas of CRD-registered kinds, enforced on CR writes
         self._cr_schemas: Dict[Tuple[str, str], dict] = {}
+        # CRD kinds declaring subresources.status: main-resource writes
+        # cannot touch .status and /status cannot touch the rest (real
+        # apiserver isolation; builtin-registered kinds are exempt so
+        # fixtures keep the envtest Status().Update convenience)
+        self._status_subresource: set = set()
         # spec.nodeName index for pods: nodeName -> {(ns, name)}; keeps
         # per-node pod LISTs (the reconcile loop's hottest query) O(pods on
         # node) instead of O(all pods)
@@ -197,6 +202,8 @@ class FakeCluster:
                 schema = (ver.get("schema") or {}).get("openAPIV3Schema")
                 if schema:
                     self._cr_schemas[(api_version, kind)] = schema
+                if "status" in (ver.get("subresources") or {}):
+                    self._status_subresource.add((api_version, kind))
 
     def lookup_kind(self, api_version: str, kind: str) -> Tuple[str, bool]:
         try:
@@ -320,6 +327,8 @@ class FakeCluster:
             bucket = self._bucket(api_version, kind)
             if (ns, name_) in bucket:
                 raise AlreadyExistsError(f"{kind} {ns}/{name_} already exists")
+            if (api_version, kind) in self._status_subresource:
+                obj.pop("status", None)  # real apiservers drop it on create
             self._validate_cr(api_version, kind, obj)
             # respect a caller-provided uid (snapshot load, fixtures with
             # pre-wired ownerReferences); assign one otherwise
@@ -461,6 +470,12 @@ class FakeCluster:
                     f"{kind} {ns}/{name_}: resourceVersion {rv} is stale"
                 )
             self._validate_cr(api_version, kind, obj)
+            if (api_version, kind) in self._status_subresource:
+                # main-resource writes cannot change status
+                if "status" in stored:
+                    obj["status"] = meta.deep_copy(stored["status"])
+                else:
+                    obj.pop("status", None)
             # Immutable server-side fields carry over.
             obj["metadata"]["uid"] = stored["metadata"]["uid"]
             obj["metadata"]["creationTimestamp"] = stored["metadata"]["creationTimestamp"]
@@ -505,6 +520,9 @@ class FakeCluster:
                 raise ConflictError(
                     f"{kind} {ns}/{name}: resourceVersion {patch_rv} is stale"
                 )
+            if (api_version, kind) in self._status_subresource and \
+                    isinstance(patch, dict) and "status" in patch:
+                patch = {k: v for k, v in patch.items() if k != "status"}
             old_node = stored.get("spec", {}).get("nodeName", "") if kind == "Pod" else None
             if (api_version, kind) in self._cr_schemas:
                 preview = meta.deep_copy(stored)
@@ -518,6 +536,34 @@ class FakeCluster:
                 self._index_pod(stored)
             if self._finalize_if_ready(api_version, kind, ns, name):
                 return meta.deep_copy(stored)
+            self._notify("MODIFIED", stored)
+            return meta.deep_copy(stored)
+
+    def patch_status(
+        self,
+        api_version: str,
+        kind: str,
+        name: str,
+        status: K8sObject,
+        namespace: str = "",
+    ) -> K8sObject:
+        """The /status subresource: merges ONLY ``.status`` (and, for kinds
+        with the subresource declared, is the only way to change it)."""
+        with self._lock:
+            _, namespaced = self.lookup_kind(api_version, kind)
+            ns = namespace if namespaced else ""
+            bucket = self._bucket(api_version, kind)
+            stored = bucket.get((ns, name))
+            if stored is None:
+                raise NotFoundError(f"{kind} {ns}/{name} not found")
+            merged = meta.deep_copy(stored)
+            meta.json_merge_patch(merged, {"status": status})
+            self._validate_cr(api_version, kind, merged)
+            if "status" in merged:
+                stored["status"] = merged["status"]
+            else:
+                stored.pop("status", None)
+            stored["metadata"]["resourceVersion"] = self._next_rv()
             self._notify("MODIFIED", stored)
             return meta.deep_copy(stored)
 

@@ -594,3 +594,63 @@ class TestCrValidationConformance:
         got = k8s.patch(f"{group}/v1", "Gauge", "g1",
                         {"spec": {"size": 9}}, "default")
         assert got["spec"]["size"] == 9
+
+
+class TestStatusSubresourceIsolation:
+    def test_crd_status_subresource_isolated(self, k8s, tmp_path):
+        """A CRD declaring subresources.status gets real-apiserver
+        isolation: main-resource writes cannot change .status and /status
+        cannot change the rest."""
+        import yaml
+
+        from k8s_operator_libs_amd.crdutil import CRD_OPERATION_APPLY, process_crds
+
+        group = f"sub{uuid.uuid4().hex[:6]}.amd.com"
+        crd = {
+            "apiVersion": "apiextensions.k8s.io/v1",
+            "kind": "CustomResourceDefinition",
+            "metadata": {"name": f"jobs.{group}"},
+            "spec": {
+                "group": group, "scope": "Namespaced",
+                "names": {"kind": "Job2", "plural": "jobs",
+                          "singular": "job2"},
+                "versions": [{
+                    "name": "v1", "served": True, "storage": True,
+                    "subresources": {"status": {}},
+                    "schema": {"openAPIV3Schema": {
+                        "type": "object",
+                        "properties": {
+                            "spec": {"type": "object",
+                                     "properties": {"size": {"type": "integer"}}},
+                            "status": {"type": "object",
+                                       "properties": {"phase": {"type": "string"}}},
+                        },
+                    }},
+                }],
+            },
+        }
+        path = tmp_path / "job-crd.yaml"
+        path.write_text(yaml.safe_dump(crd))
+        process_crds(k8s, [str(path)], CRD_OPERATION_APPLY)
+        k8s.register_kind(f"{group}/v1", "Job2", "jobs", True)
+
+        created = k8s.create({
+            "apiVersion": f"{group}/v1", "kind": "Job2",
+            "metadata": {"name": "j1", "namespace": "default"},
+            "spec": {"size": 1},
+            "status": {"phase": "sneaky"},  # dropped on create
+        })
+        assert "status" not in created or not created["status"]
+        # main patch cannot set status
+        k8s.patch(f"{group}/v1", "Job2", "j1",
+                  {"spec": {"size": 2}, "status": {"phase": "sneaky2"}},
+                  "default")
+        got = k8s.get(f"{group}/v1", "Job2", "j1", "default")
+        assert got["spec"]["size"] == 2
+        assert not got.get("status")
+        # /status sets ONLY status
+        k8s.patch_status(f"{group}/v1", "Job2", "j1", {"phase": "Ready"},
+                         "default")
+        got = k8s.get(f"{group}/v1", "Job2", "j1", "default")
+        assert got["status"]["phase"] == "Ready"
+        assert got["spec"]["size"] == 2
