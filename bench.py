@@ -321,7 +321,10 @@ def _run_gpu_validation(device, mode="inplace"):
     err = native.mfma_f32_check(device)
     if err > 1e-6:
         raise RuntimeError(f"MFMA validation failed: err={err}")
-    return err
+    bf16_err = native.mfma_bf16_check(device)
+    if bf16_err > 5e-2:
+        raise RuntimeError(f"bf16 MFMA validation failed: err={bf16_err}")
+    return max(err, bf16_err)
 
 
 def _build_substrate(substrate):
