@@ -482,7 +482,16 @@ class FakeCluster:
             if "deletionTimestamp" in stored["metadata"]:
                 obj["metadata"]["deletionTimestamp"] = stored["metadata"]["deletionTimestamp"]
             obj["metadata"]["resourceVersion"] = self._next_rv()
-            obj["metadata"]["generation"] = stored["metadata"].get("generation", 1) + 1
+            # generation bumps only on non-status content changes (real
+            # apiserver semantics for resources with a status subresource;
+            # harmless approximation for the rest)
+            def _gen_view(o):
+                return {k: v for k, v in o.items()
+                        if k not in ("metadata", "status")}
+            gen = stored["metadata"].get("generation", 1)
+            if _gen_view(obj) != _gen_view(stored):
+                gen += 1
+            obj["metadata"]["generation"] = gen
             self._index_pod(stored, remove=True)
             bucket[(ns, name_)] = obj
             self._index_pod(obj)
