@@ -200,6 +200,22 @@ def main(argv=None) -> int:
         manager, args.namespace, driver_labels, policy,
         resync_seconds=max(args.interval, 1.0),
     )
+
+    # graceful termination: SIGTERM/SIGINT stop the reconcile loop, which
+    # releases the Lease on the way out so another replica takes over
+    # immediately instead of waiting out the lease duration
+    import signal
+
+    def _shutdown(signum, frame):
+        log.info("received signal %d: shutting down", signum)
+        controller.stop()
+
+    try:
+        signal.signal(signal.SIGTERM, _shutdown)
+        signal.signal(signal.SIGINT, _shutdown)
+    except ValueError:
+        pass  # not the main thread (tests drive main() from a worker)
+
     if args.no_leader_election:
         controller.run()
     else:

@@ -5,10 +5,18 @@ import httpx
 import examples.amdgpu_upgrade_operator as operator
 
 
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def test_demo_completes_and_serves_metrics():
     rc = operator.main([
         "--demo", "--demo-nodes", "3", "--interval", "0.02",
-        "--metrics-port", "18877",
+        "--metrics-port", str(_free_port()),
     ])
     assert rc == 0
 
@@ -17,12 +25,13 @@ def test_metrics_endpoint_serves_prometheus(client):
     from k8s_operator_libs_amd.upgrade.state_manager import ClusterUpgradeStateManager
 
     manager = ClusterUpgradeStateManager(client)
-    server = operator.serve_metrics(manager, 18878)
+    port = _free_port()
+    server = operator.serve_metrics(manager, port)
     try:
-        resp = httpx.get("http://127.0.0.1:18878/metrics")
+        resp = httpx.get(f"http://127.0.0.1:{port}/metrics")
         assert resp.status_code == 200
         assert "amd_upgrade_reconcile_duration_seconds" in resp.text
-        assert httpx.get("http://127.0.0.1:18878/other").status_code == 404
+        assert httpx.get(f"http://127.0.0.1:{port}/other").status_code == 404
     finally:
         server.shutdown()
 
@@ -87,7 +96,7 @@ class TestPolicyWebhook:
 def test_demo_requestor_mode_completes():
     rc = operator.main([
         "--demo", "--demo-requestor", "--demo-nodes", "2",
-        "--interval", "0.02", "--metrics-port", "18879",
+        "--interval", "0.02", "--metrics-port", str(_free_port()),
     ])
     assert rc == 0
 
@@ -167,3 +176,52 @@ def test_production_entrypoint_with_leader_election(monkeypatch):
     finally:
         handle.stop()
         t.join(timeout=10)
+
+
+def test_sigterm_releases_lease(monkeypatch):
+    """Graceful shutdown: SIGTERM stops the controller and the elector
+    releases the Lease, so a standby replica can take over immediately."""
+    import os
+    import signal
+    import threading
+    import time
+
+    from k8s_operator_libs_amd.core.apiserver import start_apiserver
+
+    handle = start_apiserver()
+    monkeypatch.setenv("KUBERNETES_MASTER", handle.url)
+    monkeypatch.delenv("KUBECONFIG", raising=False)
+
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+
+    # run main() on the MAIN thread so signal handlers install; drive the
+    # SIGTERM from a helper thread once the lease is held
+    def killer():
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            leases = handle.cluster.list("coordination.k8s.io/v1", "Lease")
+            if leases:
+                os.kill(os.getpid(), signal.SIGTERM)
+                return
+            time.sleep(0.05)
+
+    t = threading.Thread(target=killer, daemon=True)
+    t.start()
+    old_term = signal.getsignal(signal.SIGTERM)
+    old_int = signal.getsignal(signal.SIGINT)
+    try:
+        rc = operator.main([
+            "--namespace", "amd-gpu-operator",
+            "--interval", "0.05", "--metrics-port", str(port),
+        ])
+        assert rc == 0
+        lease = handle.cluster.list("coordination.k8s.io/v1", "Lease")[0]
+        assert not lease["spec"].get("holderIdentity"), lease["spec"]
+    finally:
+        signal.signal(signal.SIGTERM, old_term)
+        signal.signal(signal.SIGINT, old_int)
+        handle.stop()
