@@ -654,3 +654,38 @@ class TestStatusSubresourceIsolation:
         got = k8s.get(f"{group}/v1", "Job2", "j1", "default")
         assert got["status"]["phase"] == "Ready"
         assert got["spec"]["size"] == 2
+
+
+class TestNegativePaths:
+    def test_get_patch_delete_absent_404(self, k8s, ns):
+        for op in ("get", "patch", "delete"):
+            with pytest.raises(NotFoundError):
+                if op == "get":
+                    k8s.get("v1", "Pod", "absent", ns)
+                elif op == "patch":
+                    k8s.patch("v1", "Pod", "absent", {"metadata": {}}, ns)
+                else:
+                    k8s.delete("v1", "Pod", "absent", ns)
+
+    def test_duplicate_create_conflicts(self, k8s, ns):
+        from k8s_operator_libs_amd.core.errors import AlreadyExistsError
+
+        name = make_pod(k8s, ns)
+        with pytest.raises(AlreadyExistsError):
+            k8s.create({"apiVersion": "v1", "kind": "Pod",
+                        "metadata": {"name": name, "namespace": ns},
+                        "spec": {"containers": [{"name": "c", "image": "x"}]}})
+
+    def test_invalid_continue_token_400(self, k8s, ns):
+        make_pod(k8s, ns)
+        resp = httpx.get(
+            f"{k8s.base_url}/api/v1/namespaces/{ns}/pods",
+            params={"limit": "1", "continue": "not-base64!@#"},
+            headers=dict(k8s._http.headers), verify=False,
+        )
+        assert resp.status_code == 400
+
+    def test_unknown_resource_404ish(self, k8s):
+        resp = httpx.get(f"{k8s.base_url}/api/v1/flurbs",
+                         headers=dict(k8s._http.headers), verify=False)
+        assert resp.status_code >= 400

@@ -422,3 +422,37 @@ class TestUvicornEngineParity:
         rest.evict_pod("uvp-0", "default")
         with pytest.raises(Exception):
             rest.get("v1", "Pod", "uvp-0", "default")
+
+
+def test_informer_recovers_from_in_stream_410_over_http(rest, server):
+    """Over the wire: the informer's watch drops, the resume window has
+    been evicted (tiny history ring), reconnect gets the in-stream
+    ERROR-410 — the informer must relist and converge."""
+    cluster = server.cluster
+    old_hist = cluster.WATCH_HISTORY
+    rest.create(node("w410"))
+    cached = CachedClient(rest)
+    try:
+        assert cached.get("v1", "Node", "w410")
+        inf = cached._informers[("v1", "Node")]
+        # shrink the ring, then push enough events to evict the resume point
+        with cluster._lock:
+            import collections as _c
+
+            key = ("v1", "Node")
+            cluster._history[key] = _c.deque(cluster._history.get(key, ()),
+                                             maxlen=4)
+        inf._watch.stop()
+        for i in range(10):
+            rest.patch("v1", "Node", "w410",
+                       {"metadata": {"labels": {"i": str(i)}}})
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if cached.get("v1", "Node", "w410")["metadata"].get(
+                    "labels", {}).get("i") == "9":
+                break
+            time.sleep(0.05)
+        assert cached.get("v1", "Node", "w410")["metadata"]["labels"]["i"] == "9"
+    finally:
+        cached.stop()
+        cluster.WATCH_HISTORY = old_hist
