@@ -778,10 +778,24 @@ class FakeCluster:
         for k in snapshot.get("kinds", []):
             cluster.register_kind(k["apiVersion"], k["kind"], k["plural"],
                                   k["namespaced"])
-        for obj in snapshot.get("objects", []):
+        # CRDs first so schema/subresource registration precedes their CRs;
+        # then restore statuses through the subresource for isolated kinds
+        # (create strips them, like a real apiserver)
+        objs = sorted(
+            snapshot.get("objects", []),
+            key=lambda o: 0 if meta.kind(o) == "CustomResourceDefinition" else 1,
+        )
+        for obj in objs:
             obj = meta.deep_copy(obj)
             obj.get("metadata", {}).pop("resourceVersion", None)
-            cluster.create(obj)
+            status = obj.get("status")
+            created = cluster.create(obj)
+            key = (meta.api_version(obj), meta.kind(obj))
+            if status is not None and key in cluster._status_subresource:
+                cluster.patch_status(
+                    meta.api_version(obj), meta.kind(obj),
+                    meta.name(created), status, meta.namespace(created),
+                )
         return cluster
 
     # -- convenience for tests/benchmarks ------------------------------------
