@@ -151,6 +151,8 @@ def create_app(cluster: Optional[FakeCluster] = None):
                 resource_version=resource_version,
                 label_selector=label_selector,
                 field_selector=qp.get("fieldSelector", ""),
+                send_initial_events=qp.get("sendInitialEvents")
+                in ("true", "1"),
             )
         except GoneError as exc:
             # expired before the stream even opened: stream a single ERROR
@@ -556,6 +558,8 @@ def _make_threaded_server(host, port, cluster):
                     resource_version=qp.get("resourceVersion") or None,
                     label_selector=qp.get("labelSelector", ""),
                     field_selector=qp.get("fieldSelector", ""),
+                    send_initial_events=qp.get("sendInitialEvents")
+                    in ("true", "1"),
                 )
             except _Gone as exc:
                 # 200 + in-stream ERROR Status, the real apiserver shape

@@ -35,6 +35,13 @@ class _Informer:
     BACKOFF_BASE = 0.1
     BACKOFF_MAX = 5.0
 
+    #: bootstrap via the WatchList protocol (KEP-3157: sendInitialEvents
+    #: streams the initial state through the watch, ending with a BOOKMARK
+    #: annotated k8s.io/initial-events-end) instead of a separate LIST —
+    #: one less round trip and no list-sized memory spike.  Falls back to
+    #: LIST-then-WATCH(rv) automatically when the server rejects it.
+    USE_WATCH_LIST = True
+
     def __init__(self, delegate: Client, api_version: str, kind: str,
                  sync_delay: float = 0.0) -> None:
         self.api_version = api_version
@@ -77,15 +84,60 @@ class _Informer:
             self.api_version, self.kind, resource_version=self._last_rv
         )
 
+    def _bootstrap_watch_list(self) -> bool:
+        """WatchList bootstrap: consume initial ADDEDs until the
+        initial-events-end bookmark, then keep the same stream live.
+        Returns False (cleanly) when the delegate/server lacks support."""
+        try:
+            w = self._delegate.watch(self.api_version, self.kind,
+                                     send_initial_events=True)
+        except TypeError:
+            return False  # delegate without the parameter
+        except Exception:
+            return False  # server rejected sendInitialEvents
+        store: Dict[Tuple[str, str], K8sObject] = {}
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline and not self._stop.is_set():
+            item = w.next(timeout=0.5)
+            if item is None:
+                alive = getattr(w, "alive", None)
+                if alive is not None and not alive():
+                    break
+                continue
+            event_type, obj = item
+            if event_type == "BOOKMARK":
+                md = (obj or {}).get("metadata", {})
+                if (md.get("annotations") or {}).get(
+                        "k8s.io/initial-events-end") == "true":
+                    with self._lock:
+                        self._store = store
+                        self._last_rv = md.get("resourceVersion")
+                        self._changed.notify_all()
+                    self._watch = w
+                    return True
+                continue
+            if event_type == "ERROR":
+                break
+            key = (meta.namespace(obj), meta.name(obj))
+            if event_type == "DELETED":
+                store.pop(key, None)
+            else:
+                store[key] = obj
+        w.stop()
+        return False
+
     def _run(self) -> None:
         from .errors import GoneError
 
         backoff = self.BACKOFF_BASE
-        # reflector bootstrap: LIST (capture list RV) -> WATCH(from that RV).
-        # With an RV-anchored delegate nothing can be lost in between; with a
-        # live-only delegate, watch-then-relist covers the gap instead.
+        # reflector bootstrap: WatchList when available, else LIST (capture
+        # list RV) -> WATCH(from that RV).  With an RV-anchored delegate
+        # nothing can be lost in between; with a live-only delegate,
+        # watch-then-relist covers the gap instead.
         while not self._stop.is_set():
             try:
+                if self.USE_WATCH_LIST and self._bootstrap_watch_list():
+                    break
                 self._relist()
                 if self._last_rv is None:
                     # legacy live-only delegate: open watch first, then

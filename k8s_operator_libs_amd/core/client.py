@@ -168,11 +168,12 @@ class FakeClient(Client):
                                          namespace)
 
     def watch(self, api_version, kind, namespace=None, resource_version=None,
-              label_selector="", field_selector=""):
+              label_selector="", field_selector="", send_initial_events=False):
         return self.cluster.watch(
             api_version, kind, namespace=namespace,
             resource_version=resource_version, label_selector=label_selector,
             field_selector=field_selector,
+            send_initial_events=send_initial_events,
         )
 
     def list_with_meta(self, api_version, kind, namespace=None,

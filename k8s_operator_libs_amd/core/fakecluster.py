@@ -697,6 +697,7 @@ class FakeCluster:
         resource_version: Optional[str] = None,
         label_selector: str = "",
         field_selector: str = "",
+        send_initial_events: bool = False,
     ) -> Watch:
         """Open a watch stream with Kubernetes resourceVersion semantics:
 
@@ -723,6 +724,23 @@ class FakeCluster:
             w = Watch(self, key, namespace=namespace,
                       label_selector=label_selector,
                       field_selector=field_selector)
+            if send_initial_events:
+                # WatchList protocol (KEP-3157, beta in 1.32): stream the
+                # current state as synthetic ADDEDs, then a BOOKMARK whose
+                # annotation k8s.io/initial-events-end marks the consistent
+                # snapshot point; live events follow losslessly (we hold
+                # the cluster lock throughout registration).
+                for obj in self._store.get(key, {}).values():
+                    w._deliver("ADDED", meta.deep_copy(obj))
+                w.events.put(("BOOKMARK", {
+                    "kind": kind, "apiVersion": api_version,
+                    "metadata": {
+                        "resourceVersion": str(self._rv_counter),
+                        "annotations": {"k8s.io/initial-events-end": "true"},
+                    },
+                }))
+                self._watches.setdefault(key, []).append(w)
+                return w
             if resource_version == "0":
                 for obj in self._store.get(key, {}).values():
                     w._deliver("ADDED", meta.deep_copy(obj))

@@ -370,7 +370,8 @@ class RestClient(Client):
         self._raise_for(resp)
 
     def watch(self, api_version: str, kind: str, namespace=None,
-              resource_version=None, label_selector="", field_selector=""):
+              resource_version=None, label_selector="", field_selector="",
+              send_initial_events=False):
         """Open a Kubernetes watch stream (``?watch=true``) with standard
         query parameters: ``resourceVersion`` anchoring, ``labelSelector``
         filtering, namespace-scoped paths, and ``allowWatchBookmarks`` so
@@ -383,7 +384,8 @@ class RestClient(Client):
         return _HttpWatch(self, api_version, kind, namespace=namespace,
                           resource_version=resource_version,
                           label_selector=label_selector,
-                          field_selector=field_selector)
+                          field_selector=field_selector,
+                          send_initial_events=send_initial_events)
 
     # -- discovery (for crdutil.wait_for_crds) ----------------------------------
 
@@ -433,7 +435,8 @@ class _HttpWatch:
 
     def __init__(self, client: RestClient, api_version: str, kind: str,
                  namespace=None, resource_version=None,
-                 label_selector="", field_selector="") -> None:
+                 label_selector="", field_selector="",
+                 send_initial_events=False) -> None:
         import queue
         import threading
 
@@ -448,6 +451,9 @@ class _HttpWatch:
             params["labelSelector"] = label_selector
         if field_selector:
             params["fieldSelector"] = field_selector
+        if send_initial_events:
+            params["sendInitialEvents"] = "true"
+            params["resourceVersionMatch"] = "NotOlderThan"
         self._error: Optional[BaseException] = None
 
         def reader():

@@ -456,3 +456,105 @@ def test_informer_recovers_from_in_stream_410_over_http(rest, server):
     finally:
         cached.stop()
         cluster.WATCH_HISTORY = old_hist
+
+
+class TestWatchListProtocol:
+    """KEP-3157 WatchList (sendInitialEvents, beta in K8s 1.32): initial
+    state streams through the watch, ending with a BOOKMARK annotated
+    k8s.io/initial-events-end; informers bootstrap without a LIST."""
+
+    def test_cluster_watchlist_semantics(self):
+        c = FakeCluster()
+        c.create(node("n1", {"a": "1"}))
+        c.create(node("n2"))
+        w = c.watch("v1", "Node", send_initial_events=True)
+        got = {}
+        end_rv = None
+        for _ in range(10):
+            etype, obj = w.next(0.5)
+            if etype == "BOOKMARK":
+                ann = obj["metadata"].get("annotations") or {}
+                assert ann.get("k8s.io/initial-events-end") == "true"
+                end_rv = obj["metadata"]["resourceVersion"]
+                break
+            assert etype == "ADDED"
+            got[obj["metadata"]["name"]] = obj
+        assert set(got) == {"n1", "n2"}
+        assert end_rv == c.current_rv()
+        # stream continues live past the marker, losslessly
+        c.patch("v1", "Node", "n1", {"metadata": {"labels": {"a": "2"}}})
+        ev = w.next(0.5)
+        assert ev[0] == "MODIFIED" and ev[1]["metadata"]["labels"]["a"] == "2"
+        w.stop()
+
+    def test_watchlist_over_http(self, rest, server):
+        rest.create(node("wl1"))
+        w = rest.watch("v1", "Node", send_initial_events=True)
+        try:
+            seen_added = False
+            for _ in range(10):
+                etype, obj = w.next(5.0)
+                if etype == "ADDED" and obj["metadata"]["name"] == "wl1":
+                    seen_added = True
+                if etype == "BOOKMARK" and (obj["metadata"].get("annotations")
+                                            or {}).get("k8s.io/initial-events-end"):
+                    break
+            assert seen_added
+        finally:
+            w.stop()
+
+    def test_informer_bootstraps_via_watchlist(self, rest, server):
+        rest.create(node("wl2"))
+        calls = {"list": 0}
+        orig = rest.list_with_meta
+
+        def counting(*a, **kw):
+            calls["list"] += 1
+            return orig(*a, **kw)
+
+        rest.list_with_meta = counting
+        cached = CachedClient(rest)
+        try:
+            assert cached.get("v1", "Node", "wl2")["metadata"]["name"] == "wl2"
+            assert calls["list"] == 0, "informer LISTed despite WatchList"
+            inf = cached._informers[("v1", "Node")]
+            assert inf._last_rv is not None
+            # live updates flow on the same stream
+            rest.patch("v1", "Node", "wl2", {"metadata": {"labels": {"x": "1"}}})
+            deadline = time.monotonic() + 5
+            while time.monotonic() < deadline:
+                if cached.get("v1", "Node", "wl2")["metadata"].get(
+                        "labels", {}).get("x") == "1":
+                    break
+                time.sleep(0.02)
+            assert cached.get("v1", "Node", "wl2")["metadata"]["labels"]["x"] == "1"
+        finally:
+            rest.list_with_meta = orig
+            cached.stop()
+
+    def test_informer_falls_back_when_server_lacks_watchlist(self):
+        """A delegate that rejects sendInitialEvents (pre-1.27 apiserver)
+        still syncs via LIST-then-WATCH."""
+        cluster = FakeCluster()
+        cluster.create(node("fb1"))
+        inner = FakeClient(cluster)
+
+        class NoWatchList(FakeClient):
+            def __init__(self):
+                self.cluster = cluster
+
+            def watch(self, api_version, kind, namespace=None,
+                      resource_version=None, label_selector="",
+                      field_selector="", send_initial_events=False):
+                if send_initial_events:
+                    raise RuntimeError("sendInitialEvents is not allowed")
+                return inner.watch(api_version, kind, namespace=namespace,
+                                   resource_version=resource_version,
+                                   label_selector=label_selector,
+                                   field_selector=field_selector)
+
+        cached = CachedClient(NoWatchList())
+        try:
+            assert cached.get("v1", "Node", "fb1")["metadata"]["name"] == "fb1"
+        finally:
+            cached.stop()
