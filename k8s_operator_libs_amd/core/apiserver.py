@@ -464,19 +464,21 @@ def _make_threaded_server(host, port, cluster):
             (api_version, plural, namespace, name, subresource, query)."""
             parsed = urllib.parse.urlsplit(self.path)
             qp = dict(urllib.parse.parse_qsl(parsed.query))
+            from .errors import NotFoundError
+
             parts = [p for p in parsed.path.split("/") if p]
             if not parts:
-                raise ApiError("not found")
+                raise NotFoundError("not found")
             if parts[0] == "api":
                 api_version = parts[1] if len(parts) > 1 else ""
                 rest = parts[2:]
             elif parts[0] == "apis":
                 if len(parts) < 3:
-                    raise ApiError("not found")
+                    raise NotFoundError("not found")
                 api_version = f"{parts[1]}/{parts[2]}"
                 rest = parts[3:]
             else:
-                raise ApiError("not found")
+                raise NotFoundError("not found")
             namespace = ""
             if rest[:1] == ["namespaces"] and len(rest) >= 2:
                 # bare namespace-object ops go through the core route

@@ -404,6 +404,14 @@ class RestClient(Client):
 
     def close(self) -> None:
         self._http.close()
+        if self._fast_netloc is not None:
+            conn = getattr(self._fast_local, "conn", None)
+            if conn is not None:
+                try:
+                    conn.close()
+                except Exception:
+                    pass
+                self._fast_local.conn = None
 
 
 class _FastResponse:
