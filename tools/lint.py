@@ -199,7 +199,10 @@ def lint_file(path: Path) -> list[str]:
     for w in caught:
         if "invalid escape sequence" in str(w.message):
             findings.append((getattr(w, "lineno", 1) or 1, "W605", str(w.message)))
-    rel = path.relative_to(REPO) if path.is_absolute() else path
+    try:
+        rel = path.relative_to(REPO) if path.is_absolute() else path
+    except ValueError:
+        rel = path  # outside the repo: report the absolute path as-is
     linter = FileLinter(rel, tree, source)
     findings.extend(linter.run())
     findings.sort()
