@@ -17,6 +17,7 @@ Configuration resolution order (``from_environment``):
 
 from __future__ import annotations
 
+import http.client as _http_client
 import json
 import os
 from typing import Any, Dict, Optional
@@ -153,6 +154,7 @@ class RestClient(Client):
                     resp = self._http.request(method, path, **kw)
             except (httpx.ConnectError, httpx.ReadError, httpx.RemoteProtocolError,
                     httpx.ConnectTimeout, httpx.ReadTimeout,
+                    _http_client.HTTPException,
                     ConnectionError, OSError) as exc:
                 if attempt >= self._retries:
                     err = ApiError(f"connection to apiserver failed: {exc}")
