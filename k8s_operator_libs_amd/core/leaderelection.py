@@ -72,10 +72,24 @@ class LeaderElector:
         self.retry_period = retry_period
         self._stop = threading.Event()
         self._leading = threading.Event()
+        # The last Lease state THIS elector wrote (from the create/patch
+        # RESPONSE).  Renewals key their optimistic lock on it instead of a
+        # fresh read: over an informer-backed client a read issued right
+        # after our own write can be STALE (the watch event has not landed
+        # yet), and renewing against the stale resourceVersion used to 409
+        # and fake-demote a healthy leader.
+        self._held_lease = None
 
     # -- lease record handling ----------------------------------------------
 
     def _try_acquire_or_renew(self) -> bool:
+        if self._held_lease is not None:
+            if self._renew_held():
+                return True
+            # fell behind (another writer touched the Lease): fall through
+            # to the read-evaluate-acquire path with a fresh view
+            self._held_lease = None
+
         try:
             lease = self.client.get(
                 LEASE_API_VERSION, LEASE_KIND, self.lease_name, self.namespace
@@ -94,7 +108,7 @@ class LeaderElector:
                 },
             }
             try:
-                self.client.create(lease)
+                self._held_lease = self.client.create(lease)
                 return True
             except AlreadyExistsError:
                 return False
@@ -116,7 +130,7 @@ class LeaderElector:
                 patch["spec"]["leaseTransitions"] = spec.get("leaseTransitions", 0) + 1
             try:
                 # optimistic lock: losing a race means someone else renewed
-                self.client.patch(
+                self._held_lease = self.client.patch(
                     LEASE_API_VERSION, LEASE_KIND, self.lease_name, patch, self.namespace
                 )
                 return True
@@ -124,9 +138,28 @@ class LeaderElector:
                 return False
         return False
 
+    def _renew_held(self) -> bool:
+        """Renew against our own last-written resourceVersion (no read)."""
+        rv = (self._held_lease.get("metadata") or {}).get("resourceVersion")
+        if not rv:
+            return False
+        try:
+            self._held_lease = self.client.patch(
+                LEASE_API_VERSION, LEASE_KIND, self.lease_name,
+                {"metadata": {"resourceVersion": rv},
+                 "spec": {"holderIdentity": self.identity,
+                          "renewTime": _now_iso()}},
+                self.namespace,
+            )
+            return True
+        except (ConflictError, NotFoundError):
+            return False
+
     def _release(self) -> None:
         try:
-            lease = self.client.get(
+            # release against the held state first (same staleness rationale
+            # as renewals); fall back to a fresh read
+            lease = self._held_lease or self.client.get(
                 LEASE_API_VERSION, LEASE_KIND, self.lease_name, self.namespace
             )
             if lease.get("spec", {}).get("holderIdentity") == self.identity:
@@ -138,6 +171,8 @@ class LeaderElector:
                 )
         except (NotFoundError, ConflictError):
             pass
+        finally:
+            self._held_lease = None
 
     # -- public API -----------------------------------------------------------
 
