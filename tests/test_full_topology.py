@@ -21,9 +21,7 @@ registry is process-global, the two operators run in SEPARATE PROCESSES —
 exactly like production — coordinating purely through the apiserver.
 """
 
-import json
 import multiprocessing
-import socket
 import time
 
 import pytest
