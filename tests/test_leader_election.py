@@ -125,3 +125,42 @@ def test_leader_election_over_rest():
     finally:
         rest.close()
         handle.stop()
+
+
+def test_renewal_survives_stale_informer_cache():
+    """Regression (found by tests/test_full_topology.py): a leader renewing
+    through an informer-backed client used to read its own just-written
+    Lease STALE, renew with the stale resourceVersion, 409 and fake-demote
+    itself.  Renewals must key on the last-written response instead."""
+    import threading
+    import time
+
+    from k8s_operator_libs_amd.core.cache import CachedClient
+    from k8s_operator_libs_amd.core.client import FakeClient
+    from k8s_operator_libs_amd.core.fakecluster import FakeCluster
+    from k8s_operator_libs_amd.core.leaderelection import LeaderElector
+
+    cluster = FakeCluster()
+    # 150 ms artificial informer lag: every read after a write is stale
+    cached = CachedClient(FakeClient(cluster), sync_delay=0.15)
+    try:
+        elector = LeaderElector(cached, "stale-cache-lease",
+                                identity="me",
+                                lease_duration=5.0, retry_period=0.05)
+        lost = threading.Event()
+        stints = []
+
+        def work():
+            # hold leadership across many renewal periods
+            t0 = time.monotonic()
+            while time.monotonic() - t0 < 1.5 and elector.is_leading():
+                time.sleep(0.05)
+            stints.append(elector.is_leading())
+            elector.stop()
+
+        elector.run(on_started_leading=work,
+                    on_stopped_leading=lost.set)
+        # ~30 renewals happened against the laggy cache; leadership held
+        assert stints == [True], "leader was fake-demoted by its own stale cache"
+    finally:
+        cached.stop()
